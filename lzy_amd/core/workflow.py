@@ -1,0 +1,132 @@
+"""Workflow context manager.
+
+Reference capability (pylzy/lzy/core/workflow.py:41-298): an active
+workflow collects lazy calls, a *barrier* flushes the queue through the
+runtime, exceptions abort the workflow, whiteboards are finalized on exit,
+and cached ops get deterministic result URIs keyed by
+op-name+version+input-hashes (workflow.py:247-281).
+"""
+from __future__ import annotations
+
+import logging
+import threading
+from typing import TYPE_CHECKING, Any, List, Optional, Sequence, Type
+
+from lzy_amd.exceptions import WorkflowAbortedError
+from lzy_amd.snapshot import DefaultSnapshot
+
+if TYPE_CHECKING:
+    from lzy_amd.core.call import LzyCall
+    from lzy_amd.core.lzy import Lzy
+
+_LOG = logging.getLogger("lzy_amd.workflow")
+
+
+class LzyWorkflow:
+    _active: threading.local = threading.local()
+
+    @classmethod
+    def get_active(cls) -> Optional["LzyWorkflow"]:
+        return getattr(cls._active, "wf", None)
+
+    def __init__(
+        self,
+        name: str,
+        owner: "Lzy",
+        env,
+        eager: bool = False,
+        interactive: bool = True,
+    ) -> None:
+        self.name = name
+        self.owner = owner
+        self.env = env
+        self.eager = eager
+        self.interactive = interactive
+        self.execution_id = ""
+        self._call_queue: List["LzyCall"] = []
+        self._started = False
+        self._finished = False
+        self._snapshot: Optional[DefaultSnapshot] = None
+        self._whiteboards: List[Any] = []  # WritableWhiteboard instances
+
+    # -- snapshot -----------------------------------------------------------
+
+    @property
+    def snapshot(self) -> DefaultSnapshot:
+        if self._snapshot is None:
+            raise RuntimeError("Workflow is not started")
+        return self._snapshot
+
+    @property
+    def call_queue(self) -> List["LzyCall"]:
+        return self._call_queue
+
+    # -- context manager ----------------------------------------------------
+
+    def __enter__(self) -> "LzyWorkflow":
+        if LzyWorkflow.get_active() is not None:
+            raise RuntimeError(
+                "Simultaneous workflows are not supported in one thread"
+            )
+        import uuid
+
+        self.execution_id = f"{self.name}-{uuid.uuid4()}"
+        self._snapshot = DefaultSnapshot(
+            serializers=self.owner.serializer_registry,
+            storage_client=self.owner.storage_client,
+            storage_uri_prefix=f"{self.owner.storage_uri}/lzy_runs/{self.execution_id}",
+            storage_name=self.owner.storage_name,
+        )
+        self.owner.runtime.start(self)
+        self._started = True
+        LzyWorkflow._active.wf = self
+        return self
+
+    def __exit__(self, exc_type, exc_val, exc_tb) -> bool:
+        try:
+            if exc_type is None:
+                self.barrier()
+                self._finalize_whiteboards()
+                self.owner.runtime.finish(self)
+            else:
+                _LOG.warning("Workflow %s aborted: %s", self.name, exc_val)
+                self.owner.runtime.abort(self)
+        finally:
+            self._finished = True
+            LzyWorkflow._active.wf = None
+        return False
+
+    # -- calls & barrier ------------------------------------------------------
+
+    def register_call(self, call: "LzyCall") -> None:
+        self._call_queue.append(call)
+        if self.eager:
+            self.barrier()
+
+    def barrier(self) -> None:
+        """Flush queued calls through the runtime (reference _barrier).
+
+        Cache-addressed result URIs (reference workflow.py:247-281) are
+        assigned at task start by the executor — by then every input is
+        materialized, so the key (op name + version + input hashes) is
+        always computable, including for chains of cached ops.
+        """
+        if not self._call_queue:
+            return
+        calls = self._call_queue
+        self._call_queue = []
+        self.owner.runtime.exec(self, calls)
+
+    # -- whiteboards ----------------------------------------------------------
+
+    def create_whiteboard(self, typ: Type, *, tags: Sequence[str] = ()) -> Any:
+        from lzy_amd.whiteboards.wb import WritableWhiteboard
+
+        wb = WritableWhiteboard(typ, tags, self)
+        self._whiteboards.append(wb)
+        return wb
+
+    def _finalize_whiteboards(self) -> None:
+        for wb in self._whiteboards:
+            wb._finalize()
+        self._whiteboards = []

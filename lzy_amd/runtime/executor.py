@@ -1,0 +1,170 @@
+"""In-process execution of one LzyCall.
+
+This is the worker-side innermost loop — the re-design of the reference's
+remote entrypoint (pylzy/lzy/api/v1/startup.py:109-185): read inputs,
+run the op, write returns, transport exceptions as first-class outputs.
+Differences by design: inputs are store references (no slot files, no
+deserialization), std-logs are captured via a thread-routing tee, and the
+op result cache is checked against the durable tier before running
+(reference CheckCache.java:31-48 semantics, done worker-side).
+"""
+from __future__ import annotations
+
+import logging
+import os
+import sys
+import time
+import traceback
+from contextlib import contextmanager
+from typing import TYPE_CHECKING, Any, Dict, Sequence, Tuple
+
+from lzy_amd.exceptions import LzyExecutionError
+from lzy_amd.proxy import is_lzy_proxy, lzy_proxy, materialize
+from lzy_amd.utils.faults import FAULTS
+from lzy_amd.utils.logs import OpLogCapture
+from lzy_amd.utils.metrics import METRICS
+
+if TYPE_CHECKING:
+    from lzy_amd.core.call import LzyCall
+
+_LOG = logging.getLogger("lzy_amd.executor")
+
+
+@contextmanager
+def _env_vars(env: Dict[str, str]):
+    old: Dict[str, Any] = {}
+    for k, v in env.items():
+        old[k] = os.environ.get(k)
+        os.environ[k] = v
+    try:
+        yield
+    finally:
+        for k, v in old.items():
+            if v is None:
+                os.environ.pop(k, None)
+            else:
+                os.environ[k] = v
+
+
+def resolve_cache_uris(call: "LzyCall") -> None:
+    """Point the call's output entries at deterministic cache URIs.
+
+    The result URI of a cached op is a pure function of op name, version
+    and input content hashes (reference workflow.py:247-281); runs at task
+    start, when every input value is materialized.
+    """
+    import hashlib
+
+    snap = call.workflow.snapshot
+    hasher = hashlib.blake2b(digest_size=16)
+    hasher.update(call.callable_name.encode())
+    hasher.update(call.version.encode())
+    for eid in call.input_entry_ids():
+        hasher.update(snap.hash_of(eid).encode())
+    key = hasher.hexdigest()
+    base = (
+        f"{call.workflow.owner.storage_uri}/lzy_cache/{call.callable_name}"
+        f"/{call.version}/{key}"
+    )
+    for i, eid in enumerate(call.entry_ids):
+        snap.update_entry_uri(eid, f"{base}/return_{i}")
+
+
+def cache_hit(call: "LzyCall") -> bool:
+    """True when every cached output blob already exists (CheckCache parity)."""
+    if not call.cache:
+        return False
+    snap = call.workflow.snapshot
+    return all(
+        snap.storage.blob_exists(snap.get_entry(eid).storage_uri)
+        for eid in call.entry_ids
+    )
+
+
+def execute_call(call: "LzyCall") -> None:
+    """Run one op in the current thread/process; raises LzyExecutionError."""
+    snap = call.workflow.snapshot
+    name = call.callable_name
+
+    FAULTS.hit("executor.before_run")
+
+    if call.cache:
+        resolve_cache_uris(call)
+    if cache_hit(call):
+        METRICS.inc("lzy_cache_hits", op=name)
+        for eid in call.entry_ids:
+            snap.load(eid)
+        return
+
+    # materialize inputs (or hand over lazy proxies when lazy_arguments)
+    args = []
+    for eid, typ in zip(call.arg_entry_ids, call.signature.arg_types):
+        args.append(_input_value(call, eid, typ))
+    kwargs = {
+        k: _input_value(call, eid, call.signature.kwarg_types.get(k, object))
+        for k, eid in call.kwarg_entry_ids.items()
+    }
+
+    capture = OpLogCapture.instance()
+    capture.route_current_thread(name)
+    t0 = time.perf_counter()
+    try:
+        with _env_vars(call.env.env_variables):
+            result = call.signature.func(*args, **kwargs)
+    except BaseException as e:  # noqa: BLE001 - transported as a value
+        elapsed = time.perf_counter() - t0
+        METRICS.observe("lzy_op_run", elapsed)
+        METRICS.inc("lzy_op_failures", op=name)
+        tb = traceback.format_exc()
+        # exception as a first-class output (reference startup.py:124-169)
+        snap.put(call.exception_id, (type(e).__name__, str(e), tb))
+        raise LzyExecutionError(
+            f"Op {name} failed: {type(e).__name__}: {e}", task_id=call.id,
+            remote_traceback=tb,
+        ) from e
+    finally:
+        capture.unroute_current_thread()
+
+    elapsed = time.perf_counter() - t0
+    METRICS.observe("lzy_op_run", elapsed)
+    METRICS.inc("lzy_op_runs", op=name)
+
+    _store_outputs(call, result)
+
+    if call.cache:
+        for eid in call.entry_ids:
+            snap.persist(eid)
+
+    FAULTS.hit("executor.after_run")
+
+
+def _input_value(call: "LzyCall", entry_id: str, typ: type) -> Any:
+    snap = call.workflow.snapshot
+    if call.lazy_arguments:
+        return lzy_proxy(entry_id, (typ,), call.workflow)
+    got = snap.try_get(entry_id)
+    if not got.found:
+        raise LzyExecutionError(
+            f"Input entry {entry_id} for op {call.callable_name} has no value",
+            task_id=call.id,
+        )
+    value = got.value
+    return materialize(value) if is_lzy_proxy(value) else value
+
+
+def _store_outputs(call: "LzyCall", result: Any) -> None:
+    snap = call.workflow.snapshot
+    n = len(call.entry_ids)
+    if n == 1:
+        outputs: Tuple[Any, ...] = (result,)
+    else:
+        if not isinstance(result, tuple) or len(result) != n:
+            raise LzyExecutionError(
+                f"Op {call.callable_name} declared {n} outputs but returned "
+                f"{type(result).__name__} of length "
+                f"{len(result) if isinstance(result, (tuple, list)) else 'n/a'}",
+                task_id=call.id,
+            )
+        outputs = result
+    for eid, value in zip(call.entry_ids, outputs):
+        snap.put(eid, materialize(value) if is_lzy_proxy(value) else value)

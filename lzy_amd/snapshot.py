@@ -1,0 +1,187 @@
+"""Workflow data snapshot: entries + tiered result store.
+
+Reference capability (pylzy/lzy/api/v1/snapshot.py:25-191): every op
+argument/return/exception is a *snapshot entry* — serialized and uploaded
+to S3 at registration, downloaded on materialization, md5-hashed for
+cache dedup.
+
+MI355X re-design: the hot tier is the in-process object store — device
+tensors stay resident in HBM (288 GB/GPU leaves room to keep whole
+pipelines resident), host objects stay as python objects; nothing is
+serialized on the hot path.  Serialization happens only on (a) explicit
+persist (result cache / whiteboard fields — durable tier on local NVMe
+via file://), or (b) cross-process channel transport (lzy_amd/channels).
+Content hashes replace the reference's md5: xxh3-class 64-bit hash via the
+C++ core when built, blake2b fallback, HIP checksum kernel for device
+tensors (lzy_amd/ops) so hashing never drags a tensor off the GPU.
+"""
+from __future__ import annotations
+
+import hashlib
+import io
+import uuid
+from dataclasses import dataclass, field
+from typing import Any, Dict, Optional, Tuple, Type
+
+from lzy_amd.serialization.registry import LzySerializerRegistry
+from lzy_amd.storage.api import StorageClient
+
+
+@dataclass
+class SnapshotEntry:
+    id: str
+    name: str
+    typ: Type
+    storage_uri: str
+    data_format: str
+
+
+@dataclass
+class TryGetResult:
+    found: bool
+    value: Any = None
+    error: Optional[BaseException] = None
+
+
+def _hash_bytes(data: bytes) -> str:
+    try:
+        from lzy_amd.sched import _core  # C++ xxhash64 when built
+
+        return format(_core.xxhash64(data), "016x")
+    except Exception:
+        return hashlib.blake2b(data, digest_size=8).hexdigest()
+
+
+def hash_value(value: Any, serializers: LzySerializerRegistry) -> str:
+    """Content hash of a value for cache keys / dedup."""
+    try:
+        import torch
+
+        if isinstance(value, torch.Tensor):
+            if value.is_cuda:
+                try:
+                    from lzy_amd.ops import device_checksum
+
+                    return format(device_checksum(value), "016x")
+                except Exception:
+                    pass
+            t = value.detach().contiguous()
+            raw = t.cpu().view(torch.uint8) if t.dtype.is_floating_point else t.cpu()
+            return _hash_bytes(
+                f"{t.dtype}|{tuple(t.shape)}".encode()
+                + raw.view(torch.uint8).numpy(force=True).tobytes()
+            )
+    except ImportError:
+        pass
+    data, _ = serializers.dumps(value)
+    return _hash_bytes(data)
+
+
+class DefaultSnapshot:
+    """Entry registry + object store for one workflow."""
+
+    def __init__(
+        self,
+        serializers: LzySerializerRegistry,
+        storage_client: StorageClient,
+        storage_uri_prefix: str,
+        storage_name: str = "default",
+    ) -> None:
+        self._serializers = serializers
+        self._storage = storage_client
+        self._prefix = storage_uri_prefix.rstrip("/")
+        self.storage_name = storage_name
+        self._entries: Dict[str, SnapshotEntry] = {}
+        self._values: Dict[str, Any] = {}
+        self._hashes: Dict[str, str] = {}
+
+    @property
+    def serializers(self) -> LzySerializerRegistry:
+        return self._serializers
+
+    @property
+    def storage(self) -> StorageClient:
+        return self._storage
+
+    # -- entries ------------------------------------------------------------
+
+    def create_entry(
+        self, name: str, typ: Type, storage_uri: Optional[str] = None
+    ) -> SnapshotEntry:
+        eid = str(uuid.uuid4())
+        ser = self._serializers.find_serializer_by_type(typ if isinstance(typ, type) else object)
+        fmt = ser.data_format() if ser else "pickle"
+        uri = storage_uri or f"{self._prefix}/{eid}.{name}"
+        entry = SnapshotEntry(id=eid, name=name, typ=typ, storage_uri=uri, data_format=fmt)
+        self._entries[eid] = entry
+        return entry
+
+    def get_entry(self, entry_id: str) -> SnapshotEntry:
+        return self._entries[entry_id]
+
+    def update_entry_uri(self, entry_id: str, uri: str) -> None:
+        self._entries[entry_id].storage_uri = uri
+
+    # -- hot tier -----------------------------------------------------------
+
+    def put(self, entry_id: str, value: Any) -> None:
+        self._values[entry_id] = value
+        self._hashes.pop(entry_id, None)
+
+    def try_get(self, entry_id: str) -> TryGetResult:
+        if entry_id in self._values:
+            return TryGetResult(True, self._values[entry_id])
+        entry = self._entries.get(entry_id)
+        if entry is not None and self._storage.blob_exists(entry.storage_uri):
+            value = self.load(entry_id)
+            return TryGetResult(True, value)
+        return TryGetResult(False)
+
+    def get(self, entry_id: str) -> Any:
+        got = self.try_get(entry_id)
+        if not got.found:
+            raise KeyError(f"No value for entry {entry_id}")
+        return got.value
+
+    def has_value(self, entry_id: str) -> bool:
+        return entry_id in self._values or (
+            entry_id in self._entries
+            and self._storage.blob_exists(self._entries[entry_id].storage_uri)
+        )
+
+    def drop_value(self, entry_id: str) -> None:
+        self._values.pop(entry_id, None)
+
+    # -- durable tier -------------------------------------------------------
+
+    def persist(self, entry_id: str) -> str:
+        """Serialize the hot value to its storage URI; returns the URI."""
+        entry = self._entries[entry_id]
+        value = self._values[entry_id]
+        data, fmt = self._serializers.dumps(value)
+        entry.data_format = fmt
+        self._storage.write_bytes(entry.storage_uri, data)
+        self._hashes[entry_id] = _hash_bytes(data)
+        return entry.storage_uri
+
+    def load(self, entry_id: str) -> Any:
+        entry = self._entries[entry_id]
+        data = self._storage.read_bytes(entry.storage_uri)
+        value = self._serializers.loads(data, entry.data_format, entry.typ)
+        self._values[entry_id] = value
+        return value
+
+    def copy_data(self, entry_id: str, to_uri: str) -> None:
+        entry = self._entries[entry_id]
+        if not self._storage.blob_exists(entry.storage_uri):
+            self.persist(entry_id)
+        self._storage.copy(entry.storage_uri, to_uri)
+
+    # -- hashing ------------------------------------------------------------
+
+    def hash_of(self, entry_id: str) -> str:
+        h = self._hashes.get(entry_id)
+        if h is None:
+            h = hash_value(self.get(entry_id), self._serializers)
+            self._hashes[entry_id] = h
+        return h

@@ -1,0 +1,149 @@
+"""Aux subsystems: metrics, fault injection, std-log capture, provisioning,
+File passing, journal-backed runtime."""
+import os
+
+import pytest
+
+from lzy_amd import Lzy, op, File
+from lzy_amd.env.provisioning import Provisioning, PoolSpec, node_pools
+from lzy_amd.exceptions import BadProvisioningError, LzyExecutionError
+from lzy_amd.utils.faults import FAULTS, InjectedFailure
+from lzy_amd.utils.metrics import METRICS, timed
+
+
+def test_metrics_counters_and_render():
+    METRICS.reset()
+    METRICS.inc("lzy_test_counter", op="foo")
+    METRICS.inc("lzy_test_counter", op="foo")
+    METRICS.set_gauge("lzy_test_gauge", 3.5)
+    with timed("lzy_test_span"):
+        pass
+    text = METRICS.render()
+    assert 'lzy_test_counter{op="foo"} 2.0' in text
+    assert "lzy_test_gauge 3.5" in text
+    assert "lzy_test_span_seconds_count 1" in text
+
+
+def test_metrics_http_endpoint():
+    import urllib.request
+
+    METRICS.reset()
+    METRICS.inc("lzy_http_metric")
+    port = METRICS.serve()
+    try:
+        body = urllib.request.urlopen(f"http://127.0.0.1:{port}/metrics").read().decode()
+        assert "lzy_http_metric 1.0" in body
+    finally:
+        METRICS.stop()
+
+
+def test_op_metrics_recorded(lzy):
+    METRICS.reset()
+
+    @op
+    def m(x: int) -> int:
+        return x
+
+    with lzy.workflow("wf"):
+        int(m(1))
+    stats = METRICS.timing_stats("lzy_op_run")
+    assert stats["count"] == 1
+    dispatch = METRICS.timing_stats("lzy_dispatch")
+    assert dispatch["count"] == 1
+
+
+def test_fault_injection_raise(lzy):
+    @op
+    def f(x: int) -> int:
+        return x
+
+    FAULTS.arm("executor.before_run")
+    try:
+        with pytest.raises(InjectedFailure):
+            with lzy.workflow("wf"):
+                int(f(1))
+    finally:
+        FAULTS.clear()
+
+
+def test_fault_countdown():
+    FAULTS.arm("p", countdown=2)
+    FAULTS.hit("p")
+    FAULTS.hit("p")
+    with pytest.raises(InjectedFailure):
+        FAULTS.hit("p")
+    assert not FAULTS.armed("p")
+
+
+def test_log_capture_prefix(lzy, capsys):
+    @op
+    def noisy(x: int) -> int:
+        print("hello from op")
+        return x
+
+    with lzy.workflow("wf"):
+        int(noisy(1))
+    out = capsys.readouterr().out
+    assert "[LZY-" in out
+    assert "hello from op" in out
+
+
+def test_env_variables_applied(lzy):
+    @op
+    def read_env() -> str:
+        return os.environ.get("LZY_TEST_VAR", "missing")
+
+    with lzy.workflow("wf") as wf:
+        r = read_env.with_env_variables(LZY_TEST_VAR="set-by-env")()
+        assert str(r) == "set-by-env"
+    assert os.environ.get("LZY_TEST_VAR") is None
+
+
+def test_provisioning_resolution():
+    pools = node_pools(8)
+    p = Provisioning(gpu_count=4)
+    assert p.resolve_pool(pools).gpu_count >= 4
+    assert Provisioning().resolve_pool(pools).gpu_count == 0
+    assert Provisioning(gpu_type="MI355X").effective_gpu_count == 1
+    with pytest.raises(BadProvisioningError):
+        Provisioning(gpu_count=16).resolve_pool(pools)
+
+
+def test_provisioning_combine():
+    base = Provisioning(cpu_count=4, gpu_count=1)
+    override = Provisioning(gpu_count=8)
+    merged = base.combine(override)
+    assert merged.gpu_count == 8
+    assert merged.cpu_count == 4
+
+
+def test_gpu_op_rejected_without_gpu(lzy):
+    import torch
+
+    if torch.cuda.is_available():
+        pytest.skip("GPU present")
+
+    @op(gpu_count=1)
+    def g(x: int) -> int:
+        return x
+
+    with pytest.raises(BadProvisioningError):
+        with lzy.workflow("wf"):
+            int(g(1))
+
+
+def test_file_passing(lzy):
+    @op
+    def make_file(text: str) -> File:
+        f = File.create_tmp()
+        f.write_text(text)
+        return f
+
+    @op
+    def read_file(f: File) -> str:
+        return f.read_text()
+
+    with lzy.workflow("wf"):
+        f = make_file("payload")
+        s = read_file(f)
+        assert str(s) == "payload"

@@ -1,0 +1,173 @@
+"""Phase-0 workflow semantics (reference scenarios: complex_graph, exec_fail).
+
+Covers: @op capture, laziness, barrier, multi-output, fan-out/fan-in DAG,
+exception transport, ops outside workflows.
+"""
+import pytest
+
+from lzy_amd import Lzy, op, materialize, is_lzy_proxy
+from lzy_amd.exceptions import LzyExecutionError
+
+
+@op
+def inc(x: int) -> int:
+    return x + 1
+
+
+@op
+def add(a: int, b: int) -> int:
+    return a + b
+
+
+@op
+def split(x: int) -> (int, int):
+    return x, x + 10
+
+
+def test_single_identity_op(lzy):
+    @op
+    def ident(x: int) -> int:
+        return x
+
+    with lzy.workflow("wf") as wf:
+        y = ident(42)
+        assert is_lzy_proxy(y)
+        assert int(y) == 42
+
+
+def test_chain(lzy):
+    with lzy.workflow("wf") as wf:
+        a = inc(1)
+        b = inc(a)
+        c = inc(b)
+        assert int(c) == 4
+
+
+def test_fan_out_fan_in(lzy):
+    with lzy.workflow("wf") as wf:
+        xs = [inc(i) for i in range(8)]
+        total = add(add(xs[0], xs[1]), add(xs[2], xs[3]))
+        assert int(total) == (1 + 2 + 3 + 4)
+
+
+def test_multi_output(lzy):
+    with lzy.workflow("wf") as wf:
+        lo, hi = split(5)
+        assert int(lo) == 5
+        assert int(hi) == 15
+
+
+def test_tuple_annotation(lzy):
+    from typing import Tuple
+
+    @op
+    def pair(x: int) -> Tuple[int, str]:
+        return x, str(x)
+
+    with lzy.workflow("wf"):
+        a, b = pair(7)
+        assert int(a) == 7
+        assert str(b) == "7"
+
+
+def test_materialize_after_exit(lzy):
+    with lzy.workflow("wf") as wf:
+        y = inc(10)
+    # proxy still materializable after workflow end
+    assert int(y) == 11
+
+
+def test_op_outside_workflow_runs_directly():
+    assert inc(5) == 6
+
+
+def test_missing_return_annotation():
+    with pytest.raises(TypeError):
+        @op
+        def bad(x):
+            return x
+
+
+def test_exception_transport(lzy):
+    @op
+    def boom(x: int) -> int:
+        raise ValueError("broken op")
+
+    with pytest.raises(LzyExecutionError) as ei:
+        with lzy.workflow("wf"):
+            y = boom(1)
+            int(y)  # force barrier
+    assert "broken op" in str(ei.value)
+    assert "ValueError" in ei.value.remote_traceback
+
+
+def test_downstream_cancelled_on_failure(lzy):
+    ran = []
+
+    @op
+    def fail(x: int) -> int:
+        raise RuntimeError("nope")
+
+    @op
+    def after(x: int) -> int:
+        ran.append(x)
+        return x
+
+    with pytest.raises(LzyExecutionError):
+        with lzy.workflow("wf"):
+            y = fail(1)
+            z = after(y)
+            int(z)
+    assert ran == []
+
+
+def test_eager_mode(lzy):
+    order = []
+
+    @op
+    def track(x: int) -> int:
+        order.append(x)
+        return x
+
+    with lzy.workflow("wf", eager=True):
+        track(1)
+        assert order == [1]
+        track(2)
+        assert order == [1, 2]
+
+
+def test_none_output(lzy):
+    @op
+    def nothing(x: int) -> None:
+        return None
+
+    with lzy.workflow("wf"):
+        r = nothing(1)
+        assert materialize(r) is None
+
+
+def test_proxy_kwargs(lzy):
+    @op
+    def combine(a: int, *, b: int = 0) -> int:
+        return a * 100 + b
+
+    with lzy.workflow("wf"):
+        x = inc(1)
+        y = combine(3, b=x)
+        assert int(y) == 302
+
+
+def test_independent_ops_parallel(lzy):
+    import threading
+    import time
+
+    barrier = threading.Barrier(4, timeout=10)
+
+    @op
+    def wait_all(i: int) -> int:
+        barrier.wait()
+        return i
+
+    with lzy.workflow("wf"):
+        rs = [wait_all(i) for i in range(4)]
+        assert sorted(int(r) for r in rs) == [0, 1, 2, 3]
