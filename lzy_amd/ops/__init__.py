@@ -1,46 +1,162 @@
 """HIP/CDNA4 data-plane kernels (gfx950).
 
-``_hipops`` is the in-tree HIP extension: vectorized pack/unpack with
-dtype cast, MFMA-based tensor checksum for content addressing.  On a GPU
-box the native path is mandatory — ops raise NativeExtensionMissing
-instead of silently falling back to eager torch.
+``libhipops.so`` is the in-tree HIP library (lzy_amd/ops/hipops.hip):
+vectorized pack/cast and the content-checksum kernels, launched on the
+caller's current torch HIP stream via a thin ctypes binding (no torch C++
+ABI coupling — tensors are passed as raw device pointers, torch's caching
+allocator owns all memory).
+
+On a GPU box the native path is mandatory: ops raise
+NativeExtensionMissing instead of silently falling back to eager torch.
 """
 from __future__ import annotations
 
+import ctypes
+import os
 from typing import Optional
-
-import torch
 
 from lzy_amd.exceptions import NativeExtensionMissing
 
-try:
-    from lzy_amd.ops import _hipops  # type: ignore[attr-defined]
+_LIB_PATH = os.path.join(os.path.dirname(__file__), "libhipops.so")
 
-    NATIVE = True
-except ImportError:
-    _hipops = None  # type: ignore[assignment]
-    NATIVE = False
+_lib: Optional[ctypes.CDLL] = None
+_load_error: Optional[str] = None
 
 
-def _require_native() -> None:
-    if _hipops is None:
+def _try_load() -> Optional[ctypes.CDLL]:
+    global _lib, _load_error
+    if _lib is not None:
+        return _lib
+    if not os.path.exists(_LIB_PATH):
+        _load_error = f"{_LIB_PATH} not built"
+        return None
+    try:
+        lib = ctypes.CDLL(_LIB_PATH)
+    except OSError as e:  # e.g. no ROCm runtime on a CPU-only box
+        _load_error = str(e)
+        return None
+    lib.lz_cast_copy.restype = ctypes.c_int
+    lib.lz_cast_copy.argtypes = [
+        ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p, ctypes.c_int,
+        ctypes.c_int64, ctypes.c_void_p,
+    ]
+    lib.lz_checksum.restype = ctypes.c_int
+    lib.lz_checksum.argtypes = [
+        ctypes.c_void_p, ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p,
+    ]
+    lib.lz_fill_pattern.restype = ctypes.c_int
+    lib.lz_fill_pattern.argtypes = [
+        ctypes.c_void_p, ctypes.c_int64, ctypes.c_uint64, ctypes.c_void_p,
+    ]
+    lib.lz_error_name.restype = ctypes.c_char_p
+    lib.lz_error_name.argtypes = [ctypes.c_int]
+    _lib = lib
+    return lib
+
+
+NATIVE = _try_load() is not None
+
+
+def _require_native() -> ctypes.CDLL:
+    lib = _try_load()
+    if lib is None:
         raise NativeExtensionMissing(
-            "lzy_amd.ops._hipops is not built; run `python setup.py "
-            "build_ext --inplace` (or __graft_entry__.build()) with "
-            "PYTORCH_ROCM_ARCH=gfx950"
+            f"lzy_amd HIP ops not available ({_load_error}); build with "
+            f"`python setup.py build_ext --inplace` / __graft_entry__.build() "
+            f"(hipcc --offload-arch=gfx950)"
         )
+    return lib
 
 
-def device_checksum(t: torch.Tensor) -> int:
-    """64-bit content hash of a device tensor, computed on-GPU."""
-    _require_native()
+def _check(err: int) -> None:
+    if err != 0:
+        lib = _require_native()
+        name = lib.lz_error_name(err).decode()
+        raise RuntimeError(f"hipops kernel failed: {name} ({err})")
+
+
+# dtype codes — must match LzDtype in hipops.hip
+def _dtype_code(dtype) -> int:
+    import torch
+
+    codes = {
+        torch.float32: 0,
+        torch.float16: 1,
+        torch.bfloat16: 2,
+        torch.float8_e4m3fn: 3,
+        torch.float8_e5m2: 4,
+        torch.uint8: 5,
+        torch.int32: 6,
+        torch.int64: 7,
+        torch.float64: 8,
+    }
+    if dtype not in codes:
+        raise ValueError(f"unsupported dtype for hipops: {dtype}")
+    return codes[dtype]
+
+
+def _current_stream_ptr() -> int:
+    import torch
+
+    return torch.cuda.current_stream().cuda_stream
+
+
+def cast_copy(src, dst) -> None:
+    """dst[i] = cast(src[i]); both device-contiguous, same numel.
+
+    Fused pack+dtype-cast on the current stream — the channel transport's
+    cast-on-the-wire primitive.
+    """
+    lib = _require_native()
+    if src.numel() != dst.numel():
+        raise ValueError("cast_copy: numel mismatch")
+    if not (src.is_cuda and dst.is_cuda):
+        raise ValueError("cast_copy: device tensors required")
+    s = src.detach().contiguous()
+    _check(
+        lib.lz_cast_copy(
+            ctypes.c_void_p(s.data_ptr()),
+            _dtype_code(s.dtype),
+            ctypes.c_void_p(dst.data_ptr()),
+            _dtype_code(dst.dtype),
+            s.numel(),
+            ctypes.c_void_p(_current_stream_ptr()),
+        )
+    )
+
+
+def device_checksum(t) -> int:
+    """64-bit content hash of a device tensor, computed on-GPU (no PCIe
+    round-trip for the data; 8 bytes come back)."""
+    import torch
+
+    lib = _require_native()
     if not t.is_cuda:
         raise ValueError("device_checksum requires a device tensor")
-    return _hipops.checksum(t.detach().contiguous().view(-1).view(torch.uint8))
+    flat = t.detach().contiguous().view(torch.uint8) if t.dtype != torch.uint8 \
+        else t.detach().contiguous()
+    out = torch.zeros(1, dtype=torch.int64, device=t.device)
+    _check(
+        lib.lz_checksum(
+            ctypes.c_void_p(flat.data_ptr()),
+            flat.numel() * flat.element_size(),
+            ctypes.c_void_p(out.data_ptr()),
+            ctypes.c_void_p(_current_stream_ptr()),
+        )
+    )
+    return int(out.item()) & 0xFFFFFFFFFFFFFFFF
 
 
-def cast_copy(src: torch.Tensor, dst: torch.Tensor) -> None:
-    """Fused pack+dtype-cast: dst[i] = cast(src[i]); both contiguous, same
-    element count, on the same device."""
-    _require_native()
-    _hipops.cast_copy(src.detach().contiguous(), dst)
+def fill_pattern(t, seed: int = 0) -> None:
+    """Deterministic device-side fill (tests / synthetic data)."""
+    lib = _require_native()
+    flat = t.detach().contiguous()
+    nbytes = flat.numel() * flat.element_size()
+    _check(
+        lib.lz_fill_pattern(
+            ctypes.c_void_p(flat.data_ptr()),
+            nbytes,
+            ctypes.c_uint64(seed),
+            ctypes.c_void_p(_current_stream_ptr()),
+        )
+    )

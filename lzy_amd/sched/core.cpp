@@ -1,0 +1,408 @@
+// lzy_amd native core: DAG store + ready-frontier scheduler + crash-resume
+// journal + xxhash64 content hashing.
+//
+// This is the MI355X-native collapse of the reference's graph-executor /
+// scheduler / allocator trio (reference: lzy/graph-executor-2
+// services/impl/GraphServiceImpl.java + algo/Algorithms.java:10-101, and
+// the durable-LRO journal pattern of lzy/long-running
+// OperationRunnerBase.java:27-99): one in-process C++ object instead of
+// three JVM microservices + Postgres.  Dispatch cost target: <1 us per
+// task-state transition, so the per-op scheduling overhead of an 8-stage
+// DAG is dominated by the op itself, not the framework (the reference's
+// floor is a 1 s scheduler tick).
+//
+// Build: pure pybind11 (no torch dependency) — see setup.py.
+
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include <cstdint>
+#include <cstdio>
+#include <cstring>
+#include <deque>
+#include <stdexcept>
+#include <string>
+#include <unordered_map>
+#include <vector>
+
+#include <fcntl.h>
+#include <sys/stat.h>
+#include <sys/types.h>
+#include <unistd.h>
+
+namespace py = pybind11;
+
+// ---------------------------------------------------------------------------
+// xxhash64 (XXH64, public domain algorithm; independent implementation)
+// ---------------------------------------------------------------------------
+
+static constexpr uint64_t P1 = 0x9E3779B185EBCA87ULL;
+static constexpr uint64_t P2 = 0xC2B2AE3D27D4EB4FULL;
+static constexpr uint64_t P3 = 0x165667B19E3779F9ULL;
+static constexpr uint64_t P4 = 0x85EBCA77C2B2AE63ULL;
+static constexpr uint64_t P5 = 0x27D4EB2F165667C5ULL;
+
+static inline uint64_t rotl64(uint64_t x, int r) {
+    return (x << r) | (x >> (64 - r));
+}
+
+static inline uint64_t read64(const uint8_t* p) {
+    uint64_t v;
+    std::memcpy(&v, p, 8);
+    return v;
+}
+
+static inline uint32_t read32(const uint8_t* p) {
+    uint32_t v;
+    std::memcpy(&v, p, 4);
+    return v;
+}
+
+static inline uint64_t xxh_round(uint64_t acc, uint64_t input) {
+    acc += input * P2;
+    acc = rotl64(acc, 31);
+    acc *= P1;
+    return acc;
+}
+
+static inline uint64_t xxh_merge(uint64_t acc, uint64_t val) {
+    acc ^= xxh_round(0, val);
+    acc = acc * P1 + P4;
+    return acc;
+}
+
+uint64_t xxhash64_raw(const uint8_t* data, size_t len, uint64_t seed) {
+    const uint8_t* p = data;
+    const uint8_t* end = data + len;
+    uint64_t h;
+
+    if (len >= 32) {
+        uint64_t v1 = seed + P1 + P2;
+        uint64_t v2 = seed + P2;
+        uint64_t v3 = seed;
+        uint64_t v4 = seed - P1;
+        const uint8_t* limit = end - 32;
+        do {
+            v1 = xxh_round(v1, read64(p));
+            v2 = xxh_round(v2, read64(p + 8));
+            v3 = xxh_round(v3, read64(p + 16));
+            v4 = xxh_round(v4, read64(p + 24));
+            p += 32;
+        } while (p <= limit);
+        h = rotl64(v1, 1) + rotl64(v2, 7) + rotl64(v3, 12) + rotl64(v4, 18);
+        h = xxh_merge(h, v1);
+        h = xxh_merge(h, v2);
+        h = xxh_merge(h, v3);
+        h = xxh_merge(h, v4);
+    } else {
+        h = seed + P5;
+    }
+
+    h += (uint64_t)len;
+
+    while (p + 8 <= end) {
+        h ^= xxh_round(0, read64(p));
+        h = rotl64(h, 27) * P1 + P4;
+        p += 8;
+    }
+    if (p + 4 <= end) {
+        h ^= (uint64_t)read32(p) * P1;
+        h = rotl64(h, 23) * P2 + P3;
+        p += 4;
+    }
+    while (p < end) {
+        h ^= (*p) * P5;
+        h = rotl64(h, 11) * P1;
+        ++p;
+    }
+
+    h ^= h >> 33;
+    h *= P2;
+    h ^= h >> 29;
+    h *= P3;
+    h ^= h >> 32;
+    return h;
+}
+
+// ---------------------------------------------------------------------------
+// DAG scheduler
+// ---------------------------------------------------------------------------
+
+enum class TaskState : uint8_t { Pending, Ready, Running, Done, Failed, Cancelled };
+
+static const char* state_name(TaskState s) {
+    switch (s) {
+        case TaskState::Pending: return "pending";
+        case TaskState::Ready: return "ready";
+        case TaskState::Running: return "running";
+        case TaskState::Done: return "done";
+        case TaskState::Failed: return "failed";
+        case TaskState::Cancelled: return "cancelled";
+    }
+    return "?";
+}
+
+class Dag {
+public:
+    void add_task(const std::string& id, const std::vector<std::string>& deps) {
+        if (sealed_) throw std::runtime_error("DAG is sealed");
+        if (index_.count(id)) throw std::invalid_argument("duplicate task " + id);
+        uint32_t idx = (uint32_t)tasks_.size();
+        index_.emplace(id, idx);
+        tasks_.push_back(Task{id, {}, {}, 0, TaskState::Pending});
+        pending_deps_.push_back(deps);
+    }
+
+    void seal() {
+        if (sealed_) return;
+        // resolve deps; unknown ids are entries produced before this batch
+        for (uint32_t i = 0; i < tasks_.size(); ++i) {
+            for (const auto& d : pending_deps_[i]) {
+                auto it = index_.find(d);
+                if (it == index_.end() || it->second == i) continue;
+                tasks_[i].deps.push_back(it->second);
+                tasks_[it->second].dependents.push_back(i);
+            }
+            tasks_[i].indegree = (uint32_t)tasks_[i].deps.size();
+        }
+        pending_deps_.clear();
+        check_acyclic();
+        for (auto& t : tasks_) {
+            if (t.indegree == 0) t.state = TaskState::Ready;
+        }
+        sealed_ = true;
+    }
+
+    std::vector<std::string> take_ready() {
+        std::vector<std::string> out;
+        for (auto& t : tasks_) {
+            if (t.state == TaskState::Ready) {
+                t.state = TaskState::Running;
+                out.push_back(t.id);
+            }
+        }
+        return out;
+    }
+
+    std::vector<std::string> complete(const std::string& id) {
+        Task& t = get(id);
+        t.state = TaskState::Done;
+        std::vector<std::string> newly;
+        for (uint32_t di : t.dependents) {
+            Task& d = tasks_[di];
+            if (--d.indegree == 0 && d.state == TaskState::Pending) {
+                d.state = TaskState::Ready;
+                newly.push_back(d.id);
+            }
+        }
+        return newly;
+    }
+
+    std::vector<std::string> fail(const std::string& id) {
+        Task& t = get(id);
+        t.state = TaskState::Failed;
+        std::vector<std::string> cancelled;
+        std::deque<uint32_t> stack(t.dependents.begin(), t.dependents.end());
+        while (!stack.empty()) {
+            uint32_t i = stack.back();
+            stack.pop_back();
+            Task& d = tasks_[i];
+            if (d.state == TaskState::Pending || d.state == TaskState::Ready) {
+                d.state = TaskState::Cancelled;
+                cancelled.push_back(d.id);
+                for (uint32_t dd : d.dependents) stack.push_back(dd);
+            }
+        }
+        return cancelled;
+    }
+
+    std::string state(const std::string& id) { return state_name(get(id).state); }
+
+    bool is_done() const {
+        for (const auto& t : tasks_) {
+            if (t.state != TaskState::Done && t.state != TaskState::Failed &&
+                t.state != TaskState::Cancelled)
+                return false;
+        }
+        return true;
+    }
+
+    std::unordered_map<std::string, int> counts() const {
+        std::unordered_map<std::string, int> out;
+        for (const auto& t : tasks_) out[state_name(t.state)]++;
+        return out;
+    }
+
+    size_t size() const { return tasks_.size(); }
+
+private:
+    struct Task {
+        std::string id;
+        std::vector<uint32_t> deps;
+        std::vector<uint32_t> dependents;
+        uint32_t indegree;
+        TaskState state;
+    };
+
+    Task& get(const std::string& id) {
+        auto it = index_.find(id);
+        if (it == index_.end()) throw std::invalid_argument("unknown task " + id);
+        return tasks_[it->second];
+    }
+
+    void check_acyclic() const {
+        // Kahn over a scratch indegree array
+        std::vector<uint32_t> indeg(tasks_.size());
+        for (size_t i = 0; i < tasks_.size(); ++i) indeg[i] = tasks_[i].indegree;
+        std::deque<uint32_t> frontier;
+        for (size_t i = 0; i < tasks_.size(); ++i)
+            if (indeg[i] == 0) frontier.push_back((uint32_t)i);
+        size_t seen = 0;
+        while (!frontier.empty()) {
+            uint32_t i = frontier.front();
+            frontier.pop_front();
+            ++seen;
+            for (uint32_t d : tasks_[i].dependents) {
+                if (--indeg[d] == 0) frontier.push_back(d);
+            }
+        }
+        if (seen != tasks_.size())
+            throw std::invalid_argument("cycle detected in task graph");
+    }
+
+    std::vector<Task> tasks_;
+    std::vector<std::vector<std::string>> pending_deps_;
+    std::unordered_map<std::string, uint32_t> index_;
+    bool sealed_ = false;
+};
+
+// ---------------------------------------------------------------------------
+// Journal: append-only task-state log, fsync'd per record (crash-safe).
+// Format: one JSON object per line {"t": id, "s": state, "d": detail}.
+// ---------------------------------------------------------------------------
+
+static std::string json_escape(const std::string& s) {
+    std::string out;
+    out.reserve(s.size() + 8);
+    for (char c : s) {
+        switch (c) {
+            case '"': out += "\\\""; break;
+            case '\\': out += "\\\\"; break;
+            case '\n': out += "\\n"; break;
+            case '\r': out += "\\r"; break;
+            case '\t': out += "\\t"; break;
+            default:
+                if ((unsigned char)c < 0x20) {
+                    char buf[8];
+                    std::snprintf(buf, sizeof buf, "\\u%04x", c);
+                    out += buf;
+                } else {
+                    out += c;
+                }
+        }
+    }
+    return out;
+}
+
+class Journal {
+public:
+    explicit Journal(const std::string& path) : path_(path) {
+        // mkdir -p parent
+        std::string dir = path.substr(0, path.find_last_of('/'));
+        if (!dir.empty()) {
+            std::string acc;
+            for (size_t i = 0; i < dir.size(); ++i) {
+                acc += dir[i];
+                if (dir[i] == '/' || i + 1 == dir.size()) {
+                    if (acc != "/") ::mkdir(acc.c_str(), 0755);
+                }
+            }
+        }
+        fd_ = ::open(path.c_str(), O_WRONLY | O_CREAT | O_APPEND, 0644);
+        if (fd_ < 0) throw std::runtime_error("cannot open journal " + path);
+    }
+
+    ~Journal() { close(); }
+
+    void record(const std::string& id, const std::string& state,
+                const std::string& detail = "") {
+        if (fd_ < 0) throw std::runtime_error("journal closed");
+        std::string line = "{\"t\": \"" + json_escape(id) + "\", \"s\": \"" +
+                           json_escape(state) + "\", \"d\": \"" +
+                           json_escape(detail) + "\"}\n";
+        ssize_t n = ::write(fd_, line.data(), line.size());
+        (void)n;
+        ::fsync(fd_);
+    }
+
+    void close() {
+        if (fd_ >= 0) {
+            ::close(fd_);
+            fd_ = -1;
+        }
+    }
+
+private:
+    std::string path_;
+    int fd_ = -1;
+};
+
+PYBIND11_MODULE(_core, m) {
+    m.doc() = "lzy_amd native core: DAG scheduler, journal, xxhash64";
+
+    m.def(
+        "xxhash64",
+        [](py::bytes data, uint64_t seed) {
+            char* buf;
+            Py_ssize_t len;
+            if (PyBytes_AsStringAndSize(data.ptr(), &buf, &len) != 0)
+                throw py::error_already_set();
+            uint64_t h;
+            {
+                py::gil_scoped_release rel;
+                h = xxhash64_raw((const uint8_t*)buf, (size_t)len, seed);
+            }
+            return h;
+        },
+        py::arg("data"), py::arg("seed") = 0);
+
+    py::class_<Dag>(m, "Dag")
+        .def(py::init<>())
+        .def("add_task", &Dag::add_task, py::arg("tid"), py::arg("deps"))
+        .def("seal", &Dag::seal)
+        .def("take_ready", &Dag::take_ready)
+        .def("complete", &Dag::complete, py::arg("tid"))
+        .def("fail", &Dag::fail, py::arg("tid"))
+        .def("state", &Dag::state, py::arg("tid"))
+        .def("is_done", &Dag::is_done)
+        .def("counts", &Dag::counts)
+        .def("__len__", &Dag::size);
+
+    py::class_<Journal>(m, "Journal")
+        .def(py::init<const std::string&>(), py::arg("path"))
+        .def("record", &Journal::record, py::arg("tid"), py::arg("state"),
+             py::arg("detail") = "")
+        .def("close", &Journal::close)
+        .def_static("replay", [](const std::string& path) {
+            // replay via python's json for robustness against torn tails
+            py::dict states;
+            py::object json = py::module_::import("json");
+            py::object os_path = py::module_::import("os.path");
+            if (!os_path.attr("exists")(path).cast<bool>()) return states;
+            py::object open_f = py::module_::import("builtins").attr("open");
+            py::object f = open_f(path);
+            for (py::handle line : f) {
+                std::string s = py::str(line).cast<std::string>();
+                if (s.find_first_not_of(" \t\r\n") == std::string::npos) continue;
+                try {
+                    py::dict rec = json.attr("loads")(s);
+                    states[rec["t"]] = rec["s"];
+                } catch (py::error_already_set& e) {
+                    e.restore();
+                    PyErr_Clear();  // torn tail write after a crash
+                }
+            }
+            f.attr("close")();
+            return states;
+        });
+}

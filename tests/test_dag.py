@@ -119,3 +119,9 @@ def test_xxhash64_deterministic():
     assert a == b
     assert a != c
     assert isinstance(a, int)
+
+
+def test_xxhash64_known_vectors():
+    # official XXH64 test vectors (seed 0)
+    assert xxhash64(b"") == 0xEF46DB3751D8E999
+    assert xxhash64(b"hello") == 0x26C7827D889F6DA3
