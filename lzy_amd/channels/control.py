@@ -1,0 +1,112 @@
+"""Control plane: star topology over TCP on 127.0.0.1.
+
+Re-design of the reference's control RPC mesh (lzy-service / scheduler /
+worker gRPC with JWT, util-grpc GrpcUtils.java:31-79): on one node the
+driver (rank 0) listens; every worker rank holds one full-duplex pickled
+connection to it.  Commands flow driver->worker, events worker->driver.
+Latency is tens of microseconds vs the reference's RPC+poll seconds.
+
+The listener port travels to workers through a torch.distributed
+broadcast (CPU tensor, gloo path of the default group).
+"""
+from __future__ import annotations
+
+import logging
+import threading
+from multiprocessing.connection import Client, Connection, Listener
+from typing import Any, Callable, Dict, List, Optional
+
+import torch
+import torch.distributed as dist
+
+_LOG = logging.getLogger("lzy_amd.control")
+
+_AUTHKEY = b"lzy-amd-pool"
+
+
+class DriverControl:
+    """Rank-0 side: accepts one connection per worker rank."""
+
+    def __init__(self, world_size: int, on_event: Callable[[int, dict], None]):
+        self._world = world_size
+        self._on_event = on_event
+        self._listener = Listener(("127.0.0.1", 0), authkey=_AUTHKEY)
+        self._conns: Dict[int, Connection] = {}
+        self._lock = threading.Lock()
+        self._threads: List[threading.Thread] = []
+        self._closed = False
+
+    @property
+    def port(self) -> int:
+        return self._listener.address[1]
+
+    def accept_all(self) -> None:
+        """Accept world_size connections (including rank 0's own loopback)."""
+        for _ in range(self._world):
+            conn = self._listener.accept()
+            hello = conn.recv()
+            rank = hello["rank"]
+            self._conns[rank] = conn
+            t = threading.Thread(
+                target=self._reader, args=(rank, conn), daemon=True,
+                name=f"lzy-ctrl-r{rank}",
+            )
+            t.start()
+            self._threads.append(t)
+
+    def _reader(self, rank: int, conn: Connection) -> None:
+        try:
+            while True:
+                msg = conn.recv()
+                self._on_event(rank, msg)
+        except (EOFError, OSError):
+            if not self._closed:
+                _LOG.warning("control connection to rank %d closed", rank)
+
+    def send(self, rank: int, msg: dict) -> None:
+        with self._lock:
+            self._conns[rank].send(msg)
+
+    def broadcast(self, msg: dict) -> None:
+        with self._lock:
+            for conn in self._conns.values():
+                conn.send(msg)
+
+    def close(self) -> None:
+        self._closed = True
+        with self._lock:
+            for conn in self._conns.values():
+                try:
+                    conn.close()
+                except OSError:
+                    pass
+        self._listener.close()
+
+
+class WorkerControl:
+    """Worker side: one connection to the driver."""
+
+    def __init__(self, rank: int, port: int):
+        self._conn = Client(("127.0.0.1", port), authkey=_AUTHKEY)
+        self._send_lock = threading.Lock()
+        self._conn.send({"rank": rank})
+
+    def recv(self) -> dict:
+        return self._conn.recv()
+
+    def send_event(self, msg: dict) -> None:
+        with self._send_lock:
+            self._conn.send(msg)
+
+    def close(self) -> None:
+        try:
+            self._conn.close()
+        except OSError:
+            pass
+
+
+def broadcast_port(port: Optional[int]) -> int:
+    """Rank 0 passes its listener port; everyone gets it back."""
+    t = torch.tensor([port or 0], dtype=torch.int64)
+    dist.broadcast(t, src=0)
+    return int(t.item())

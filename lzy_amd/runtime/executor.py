@@ -46,25 +46,29 @@ def _env_vars(env: Dict[str, str]):
                 os.environ[k] = v
 
 
-def resolve_cache_uris(call: "LzyCall") -> None:
-    """Point the call's output entries at deterministic cache URIs.
-
-    The result URI of a cached op is a pure function of op name, version
-    and input content hashes (reference workflow.py:247-281); runs at task
-    start, when every input value is materialized.
-    """
+def cache_base_uri(storage_uri: str, op_name: str, version: str, input_hashes) -> str:
+    """Deterministic cache location: pure function of op name, version and
+    input content hashes (reference workflow.py:247-281).  Shared by the
+    local executor and the pool workers so cache hits are rank-agnostic."""
     import hashlib
 
-    snap = call.workflow.snapshot
     hasher = hashlib.blake2b(digest_size=16)
-    hasher.update(call.callable_name.encode())
-    hasher.update(call.version.encode())
-    for eid in call.input_entry_ids():
-        hasher.update(snap.hash_of(eid).encode())
-    key = hasher.hexdigest()
-    base = (
-        f"{call.workflow.owner.storage_uri}/lzy_cache/{call.callable_name}"
-        f"/{call.version}/{key}"
+    hasher.update(op_name.encode())
+    hasher.update(version.encode())
+    for h in input_hashes:
+        hasher.update(h.encode())
+    return f"{storage_uri}/lzy_cache/{op_name}/{version}/{hasher.hexdigest()}"
+
+
+def resolve_cache_uris(call: "LzyCall") -> None:
+    """Point the call's output entries at deterministic cache URIs; runs at
+    task start, when every input value is materialized."""
+    snap = call.workflow.snapshot
+    base = cache_base_uri(
+        call.workflow.owner.storage_uri,
+        call.callable_name,
+        call.version,
+        [snap.hash_of(eid) for eid in call.input_entry_ids()],
     )
     for i, eid in enumerate(call.entry_ids):
         snap.update_entry_uri(eid, f"{base}/return_{i}")

@@ -1,0 +1,697 @@
+"""GpuPoolRuntime: one process per MI355X GPU, driven by an in-process scheduler.
+
+This is the MI355X-native collapse of the reference's entire service fleet
+(reference layers L3-L5: lzy-service ExecuteGraph pipeline, graph-executor-2
+ExecuteTaskAction, scheduler job chains, allocator VM lifecycle — SURVEY.md
+§3.2): on one 8-GPU node there are no VMs to allocate and no conda to sync,
+so "allocation" is picking a rank, "scheduling" is a ready-frontier walk of
+the C++ DAG core, and dispatch is a ~50 us control message instead of a
+1 s tick + K8s pod create.
+
+Topology:
+  * rank 0 = driver: runs the user workflow script, owns the scheduler,
+    and is ALSO a worker (its GPU is not idle).
+  * ranks 1..N-1 = workers: enter the serve loop at pool init and execute
+    tasks / transfers until shutdown (then sys.exit(0)).
+  * control plane: one TCP connection per worker to the driver (star).
+  * data plane: torch.distributed isend/irecv on a dedicated process
+    group — RCCL over xGMI for device tensors, gloo for CPU bytes;
+    transfer commands carry a driver-assigned global sequence so every
+    rank issues its p2p ops in a pairwise-consistent order (RCCL matches
+    by order, not tags — no deadlocks by construction).
+  * gang ops (@op(gpu_count=k)): the same TaskSpec dispatched to k ranks
+    with a shared RCCL subgroup for in-op collectives/DDP.
+"""
+from __future__ import annotations
+
+import atexit
+import logging
+import os
+import queue
+import sys
+import tempfile
+import threading
+import time
+import uuid
+from typing import TYPE_CHECKING, Any, Dict, List, Optional, Sequence, Set, Tuple
+
+import torch
+import torch.distributed as dist
+
+from lzy_amd.channels.control import DriverControl, WorkerControl, broadcast_port
+from lzy_amd.channels.transport import (
+    KIND_BYTES,
+    KIND_TENSOR,
+    EntryMeta,
+    Transport,
+    describe_value,
+    pickle_value,
+    unpickle_value,
+)
+from lzy_amd.exceptions import BadProvisioningError, LzyExecutionError
+from lzy_amd.runtime.base import Runtime
+from lzy_amd.runtime.taskspec import TaskResult, TaskSpec, WorkerStore, run_taskspec
+from lzy_amd.sched import Dag, Journal
+from lzy_amd.serialization.registry import LzySerializerRegistry
+from lzy_amd.storage.api import StorageConfig
+from lzy_amd.storage.fs import FsStorageClient
+from lzy_amd.utils.logs import OpLogCapture
+from lzy_amd.utils.metrics import METRICS
+
+if TYPE_CHECKING:
+    from lzy_amd.core.call import LzyCall
+    from lzy_amd.core.workflow import LzyWorkflow
+
+_LOG = logging.getLogger("lzy_amd.pool")
+
+INLINE_LIMIT = 256 << 10  # CPU values up to 256 KiB travel inside the TaskSpec
+
+
+def _storage_root() -> str:
+    root = os.environ.get("LZY_AMD_STORAGE") or os.path.join(
+        tempfile.gettempdir(), "lzy_amd_storage"
+    )
+    return root
+
+
+class WorkerAgent:
+    """Per-rank serve loop + executor thread (every rank, driver included)."""
+
+    def __init__(self, rank: int, world: int, port: int, pg_data, device):
+        self.rank = rank
+        self.world = world
+        self.device = device
+        self.store = WorkerStore()
+        self.serializers = LzySerializerRegistry()
+        self.storage = FsStorageClient()
+        self.transport = Transport(pg_data, device)
+        self.ctrl = WorkerControl(rank, port)
+        self._pending: Dict[str, Tuple[list, Any]] = {}  # entry -> (works, fin)
+        self._outbox: List[Tuple[list, Any]] = []
+        self._groups: Dict[str, Any] = {}
+        self._exec_q: "queue.Queue[dict]" = queue.Queue()
+        self._shutdown = False
+        self._exec_thread = threading.Thread(
+            target=self._executor, daemon=True, name=f"lzy-exec-r{rank}"
+        )
+        self._exec_thread.start()
+        OpLogCapture.instance().install()
+
+    # -- serve loop ---------------------------------------------------------
+
+    def serve_forever(self) -> None:
+        while not self._shutdown:
+            try:
+                msg = self.ctrl.recv()
+            except (EOFError, OSError):
+                break
+            self._handle(msg)
+
+    def _handle(self, msg: dict) -> None:
+        cmd = msg["cmd"]
+        if cmd == "task":
+            self._exec_q.put(msg)
+        elif cmd == "xfer_send":
+            eid = msg["entry"]
+            value = self.store.get(eid)
+            works, keep = self.transport.isend_value(
+                value, self.store.pickled.get(eid), msg["dst"]
+            )
+            self._outbox.append((works, keep))
+            self._prune_outbox()
+        elif cmd == "xfer_recv":
+            meta = EntryMeta.from_wire(msg["meta"])
+            works, fin = self.transport.irecv_value(meta, msg["src"])
+            self._pending[msg["entry"]] = (works, fin)
+        elif cmd == "settle":
+            self._exec_q.put(msg)
+        elif cmd == "new_group":
+            ranks = msg["ranks"]
+            tag = msg["tag"]
+            if tag not in self._groups:
+                self._groups[tag] = dist.new_group(ranks=ranks)
+            self.ctrl.send_event({"ev": "ack", "tag": tag, "rank": self.rank})
+        elif cmd == "barrier":
+            self._exec_q.put(msg)
+        elif cmd == "load_serializers":
+            self.serializers.load_user_serializers(msg["payload"])
+            self.ctrl.send_event({"ev": "ack", "tag": msg["tag"], "rank": self.rank})
+        elif cmd == "drop_entries":
+            for eid in msg["entries"]:
+                self.store.drop(eid)
+        elif cmd == "clear_store":
+            self.store.clear()
+        elif cmd == "shutdown":
+            self._shutdown = True
+            self._exec_q.put({"cmd": "_stop"})
+
+    def _prune_outbox(self) -> None:
+        still = []
+        for works, keep in self._outbox:
+            if not all(w.is_completed() for w in works):
+                still.append((works, keep))
+        self._outbox = still
+
+    # -- executor thread ----------------------------------------------------
+
+    def _executor(self) -> None:
+        while True:
+            msg = self._exec_q.get()
+            cmd = msg["cmd"]
+            if cmd == "_stop":
+                return
+            try:
+                if cmd == "task":
+                    self._run_task(msg)
+                elif cmd == "settle":
+                    self._settle(msg["entries"])
+                    self.ctrl.send_event(
+                        {"ev": "settled", "tag": msg["tag"], "rank": self.rank}
+                    )
+                elif cmd == "barrier":
+                    if self.device is not None:
+                        torch.cuda.synchronize(self.device)
+                    if dist.is_initialized():
+                        dist.barrier()
+                    if self.device is not None:
+                        torch.cuda.synchronize(self.device)
+                    self.ctrl.send_event(
+                        {
+                            "ev": "barrier_done",
+                            "tag": msg["tag"],
+                            "rank": self.rank,
+                            "ts": time.perf_counter(),
+                        }
+                    )
+            except BaseException as e:  # noqa: BLE001 - agent must survive
+                _LOG.exception("agent r%d failed handling %s", self.rank, cmd)
+                self.ctrl.send_event(
+                    {
+                        "ev": "agent_error",
+                        "rank": self.rank,
+                        "error": f"{type(e).__name__}: {e}",
+                        "task_id": msg.get("spec").task_id if msg.get("spec") else None,
+                    }
+                )
+
+    def _settle(self, entries: Sequence[str]) -> None:
+        """Complete pending transfers, landing values in the store."""
+        for eid in entries:
+            pending = self._pending.pop(eid, None)
+            if pending is None:
+                continue
+            works, fin = pending
+            for w in works:
+                w.wait()
+            value = fin()
+            pickled = None
+            if not isinstance(value, torch.Tensor):
+                pickled = self.store.pickled.get(eid)
+            self.store.put(eid, value, pickled=pickled)
+
+    def _run_task(self, msg: dict) -> None:
+        spec: TaskSpec = msg["spec"]
+        self._settle(spec.wait_entries)
+        if self.device is not None:
+            torch.cuda.set_device(self.device)
+        gang_group = None
+        if spec.gang is not None:
+            tag = spec.gang["tag"]
+            gang_group = self._groups.get(tag)  # None -> default pg
+        result = run_taskspec(
+            spec, self.store, self.serializers, self.storage, gang_group=gang_group
+        )
+        self.ctrl.send_event(
+            {
+                "ev": "task_done" if result.ok else "task_failed",
+                "rank": self.rank,
+                "result": result,
+            }
+        )
+
+
+class GpuPool:
+    """Process-wide pool singleton."""
+
+    _instance: Optional["GpuPool"] = None
+
+    def __init__(self) -> None:
+        self.rank = int(os.environ.get("RANK", "0"))
+        self.world = int(os.environ.get("WORLD_SIZE", "1"))
+        self.is_driver = self.rank == 0
+        self.device: Optional[torch.device] = None
+        self.driver_ctrl: Optional[DriverControl] = None
+        self.agent: Optional[WorkerAgent] = None
+        self.events: "queue.Queue[Tuple[int, dict]]" = queue.Queue()
+        self._acks: Dict[str, Set[int]] = {}
+        self._group_tags: Set[str] = set()
+        self._seq = 0
+
+    @classmethod
+    def get(cls) -> "GpuPool":
+        if cls._instance is None:
+            cls._instance = GpuPool()
+            cls._instance._init()
+        return cls._instance
+
+    def _init(self) -> None:
+        if torch.cuda.is_available():
+            local = int(os.environ.get("LOCAL_RANK", self.rank))
+            self.device = torch.device("cuda", local % torch.cuda.device_count())
+            torch.cuda.set_device(self.device)
+
+        pg_data = None
+        if self.world > 1:
+            if not dist.is_initialized():
+                dist.init_process_group(
+                    backend=None,
+                    rank=self.rank,
+                    world_size=self.world,
+                    device_id=self.device,
+                )
+            pg_data = dist.new_group()  # dedicated transfer group
+
+        if self.is_driver:
+            self.driver_ctrl = DriverControl(self.world, self._on_event)
+            port = self.driver_ctrl.port
+            accept_thread = threading.Thread(
+                target=self.driver_ctrl.accept_all, daemon=True, name="lzy-accept"
+            )
+            accept_thread.start()
+        else:
+            port = None
+        if self.world > 1:
+            port = broadcast_port(port)
+
+        self.agent = WorkerAgent(self.rank, self.world, port, pg_data, self.device)
+
+        if self.is_driver:
+            accept_thread.join()
+            serve_thread = threading.Thread(
+                target=self.agent.serve_forever, daemon=True, name="lzy-agent-serve"
+            )
+            serve_thread.start()
+            atexit.register(self.shutdown)
+        else:
+            # workers serve until shutdown, then exit the process; the
+            # user script body after pool entry never runs on workers.
+            self.agent.serve_forever()
+            if dist.is_initialized():
+                try:
+                    dist.destroy_process_group()
+                except Exception:
+                    pass
+            sys.exit(0)
+
+    # -- driver-side event plumbing -----------------------------------------
+
+    def _on_event(self, rank: int, msg: dict) -> None:
+        ev = msg.get("ev")
+        if ev in ("ack", "settled"):
+            self._acks.setdefault(f"{ev}:{msg['tag']}", set()).add(rank)
+        else:
+            self.events.put((rank, msg))
+
+    def wait_acks(self, kind: str, tag: str, ranks: Sequence[int], timeout: float = 300.0) -> None:
+        deadline = time.monotonic() + timeout
+        key = f"{kind}:{tag}"
+        while time.monotonic() < deadline:
+            got = self._acks.get(key, set())
+            if set(ranks).issubset(got):
+                self._acks.pop(key, None)
+                return
+            time.sleep(0.0005)
+        raise TimeoutError(f"waiting for {kind} {tag} from {ranks}")
+
+    def next_seq(self) -> int:
+        self._seq += 1
+        return self._seq
+
+    def ensure_group(self, ranks: Sequence[int]) -> str:
+        """All ranks must call dist.new_group with the same list — broadcast
+        the creation command and wait for every rank's ack."""
+        ranks = sorted(ranks)
+        if len(ranks) == self.world:
+            return "__default__"
+        tag = ",".join(map(str, ranks))
+        if tag not in self._group_tags:
+            self.driver_ctrl.broadcast({"cmd": "new_group", "ranks": ranks, "tag": tag})
+            self.wait_acks("ack", tag, range(self.world))
+            self._group_tags.add(tag)
+        return tag
+
+    def sync_all(self) -> Dict[int, float]:
+        """Barrier across all ranks (through exec queues, so it orders after
+        all dispatched tasks); returns per-rank completion timestamps."""
+        tag = f"b{self.next_seq()}"
+        self.driver_ctrl.broadcast({"cmd": "barrier", "tag": tag})
+        deadline = time.monotonic() + 600
+        ts: Dict[int, float] = {}
+        stash = []
+        while len(ts) < self.world:
+            if time.monotonic() > deadline:
+                raise TimeoutError("pool barrier timed out")
+            try:
+                rank, msg = self.events.get(timeout=1.0)
+            except queue.Empty:
+                continue
+            if msg.get("ev") == "barrier_done" and msg.get("tag") == tag:
+                ts[rank] = msg["ts"]
+            else:
+                stash.append((rank, msg))  # foreign event; re-deliver after
+        for item in stash:
+            self.events.put(item)
+        return ts
+
+    def shutdown(self) -> None:
+        if self.driver_ctrl is not None:
+            try:
+                self.driver_ctrl.broadcast({"cmd": "shutdown"})
+                time.sleep(0.2)
+                self.driver_ctrl.close()
+            except Exception:
+                pass
+            self.driver_ctrl = None
+            if dist.is_initialized():
+                try:
+                    dist.destroy_process_group()
+                except Exception:
+                    pass
+
+
+class GpuPoolRuntime(Runtime):
+    """Driver-side Runtime implementation over the pool."""
+
+    def __init__(self, journal_dir: Optional[str] = None):
+        self._journal_dir = journal_dir or os.path.join(
+            tempfile.gettempdir(), "lzy_amd_journal"
+        )
+        self._journal: Optional[Journal] = None
+        self._pool: Optional[GpuPool] = None
+
+    @property
+    def pool(self) -> GpuPool:
+        if self._pool is None:
+            self._pool = GpuPool.get()
+        return self._pool
+
+    def storage(self) -> Optional[StorageConfig]:
+        return StorageConfig(uri=f"file://{_storage_root()}")
+
+    def start(self, workflow: "LzyWorkflow") -> None:
+        pool = self.pool  # workers never get past this line (serve loop)
+        assert pool.is_driver
+        # the rank-0 agent store backs the workflow snapshot: captured args
+        # are instantly "on" rank 0, and fetched outputs appear in the
+        # snapshot without copies.
+        workflow.snapshot._values = pool.agent.store.values
+        workflow._entry_meta = {}
+
+        def _fetcher(entry_id: str) -> None:
+            meta = workflow._entry_meta.get(entry_id)
+            if meta is not None and meta.owners and 0 not in meta.owners:
+                self.fetch_entry(entry_id, meta)
+
+        workflow.snapshot.fetcher = _fetcher
+        self._journal = Journal(
+            os.path.join(self._journal_dir, f"{workflow.execution_id}.jsonl")
+        )
+        payload = workflow.owner.serializer_registry.user_serializers_payload()
+        tag = f"ser{pool.next_seq()}"
+        pool.driver_ctrl.broadcast(
+            {"cmd": "load_serializers", "payload": payload, "tag": tag}
+        )
+        pool.wait_acks("ack", tag, range(pool.world))
+
+    def exec(self, workflow: "LzyWorkflow", calls: Sequence["LzyCall"]) -> None:
+        sched = _DriverScheduler(self.pool, workflow, calls, self._journal)
+        sched.run()
+
+    def finish(self, workflow: "LzyWorkflow") -> None:
+        if self._journal is not None:
+            self._journal.close()
+            self._journal = None
+
+    def abort(self, workflow: "LzyWorkflow") -> None:
+        self.finish(workflow)
+
+    # -- entry materialization on the driver --------------------------------
+
+    def fetch_entry(self, entry_id: str, meta: EntryMeta) -> None:
+        """Pull an entry owned by another rank into rank 0's store."""
+        pool = self.pool
+        owner = next(iter(meta.owners - {0}), None)
+        if owner is None:
+            return
+        pool.driver_ctrl.send(owner, {"cmd": "xfer_send", "entry": entry_id, "dst": 0})
+        pool.driver_ctrl.send(
+            0, {"cmd": "xfer_recv", "entry": entry_id, "src": owner,
+                "meta": meta.to_wire()}
+        )
+        tag = f"f{pool.next_seq()}"
+        pool.driver_ctrl.send(0, {"cmd": "settle", "entries": [entry_id], "tag": tag})
+        pool.wait_acks("settled", tag, [0])
+        meta.owners.add(0)
+
+
+class _DriverScheduler:
+    """One barrier batch: DAG -> placement -> dispatch -> completion."""
+
+    def __init__(self, pool: GpuPool, workflow: "LzyWorkflow",
+                 calls: Sequence["LzyCall"], journal) -> None:
+        self.pool = pool
+        self.workflow = workflow
+        self.calls = {c.id: c for c in calls}
+        self.journal = journal
+        self.meta: Dict[str, EntryMeta] = getattr(workflow, "_entry_meta", {})
+        workflow._entry_meta = self.meta
+        self.outstanding: Dict[int, int] = {r: 0 for r in range(pool.world)}
+        self.task_ranks: Dict[str, List[int]] = {}
+        self.gang_pending: Dict[str, Set[int]] = {}
+        self.errors: List[BaseException] = []
+        self.inflight = 0
+
+    # -- metadata helpers ---------------------------------------------------
+
+    def _meta_for_driver_entry(self, eid: str) -> EntryMeta:
+        m = self.meta.get(eid)
+        if m is None:
+            value = self.pool.agent.store.get(eid)
+            m = describe_value(eid, value)
+            if m.kind == KIND_BYTES:
+                data = pickle_value(value)
+                self.pool.agent.store.pickled[eid] = data
+                m.nbytes = len(data)
+            m.owners = {0}
+            self.meta[eid] = m
+        return m
+
+    # -- main loop ----------------------------------------------------------
+
+    def run(self) -> None:
+        t0 = time.perf_counter()
+        producer: Dict[str, str] = {}
+        for c in self.calls.values():
+            for eid in c.entry_ids:
+                producer[eid] = c.id
+
+        dag = Dag()
+        for c in self.calls.values():
+            deps = sorted({
+                producer[eid]
+                for eid in c.input_entry_ids()
+                if eid in producer and producer[eid] != c.id
+            })
+            dag.add_task(c.id, deps)
+        dag.seal()
+        METRICS.observe("lzy_graph_build", time.perf_counter() - t0)
+
+        for tid in dag.take_ready():
+            self._dispatch(tid)
+
+        pool = self.pool
+        failed_tasks: Set[str] = set()
+        while self.inflight > 0:
+            rank, msg = pool.events.get()
+            ev = msg.get("ev")
+            if ev == "task_done":
+                result: TaskResult = msg["result"]
+                finished = self._on_done(rank, result)
+                if finished and result.task_id not in failed_tasks:
+                    for tid in dag.complete(result.task_id):
+                        self._dispatch(tid)
+            elif ev == "task_failed":
+                result = msg["result"]
+                self._on_failed(rank, result)
+                if result.task_id not in failed_tasks:
+                    failed_tasks.add(result.task_id)
+                    for ct in dag.fail(result.task_id):
+                        self.journal.record(ct, "cancelled")
+            elif ev == "agent_error":
+                self.errors.append(
+                    LzyExecutionError(f"agent rank {rank}: {msg['error']}")
+                )
+                self.inflight -= 1
+            # barrier_done etc. are routed via acks, not here
+
+        if self.errors:
+            raise self.errors[0]
+
+    # -- dispatch ------------------------------------------------------------
+
+    def _dispatch(self, task_id: str) -> None:
+        call = self.calls[task_id]
+        gpu_count = call.env.provisioning.effective_gpu_count
+        pool = self.pool
+
+        if gpu_count > pool.world:
+            self.errors.append(BadProvisioningError(
+                f"op {call.callable_name} needs gpu_count={gpu_count}, pool has "
+                f"{pool.world} ranks"
+            ))
+            return
+
+        dispatch_t0 = time.perf_counter()
+        if gpu_count > 1:
+            ranks = self._pick_gang(gpu_count)
+            tag = pool.ensure_group(ranks)
+            gang = {"ranks": ranks, "tag": tag}
+        else:
+            ranks = [self._pick_rank(call)]
+            gang = None
+
+        self.task_ranks[task_id] = ranks
+        if gang is not None:
+            self.gang_pending[task_id] = set(ranks)
+
+        specs_inline: Dict[str, bytes] = {}
+        wait_entries_per_rank: Dict[int, List[str]] = {r: [] for r in ranks}
+
+        for eid in call.input_entry_ids():
+            meta = self.meta.get(eid)
+            if meta is None:
+                # driver-captured arg (or earlier-batch result on rank 0)
+                meta = self._meta_for_driver_entry(eid)
+            if (
+                meta.kind == KIND_BYTES
+                and meta.nbytes <= INLINE_LIMIT
+                and 0 in meta.owners
+            ):
+                specs_inline[eid] = self.pool.agent.store.pickled.get(
+                    eid
+                ) or pickle_value(self.pool.agent.store.get(eid))
+                continue
+            for r in ranks:
+                if r not in meta.owners:
+                    owner = 0 if 0 in meta.owners else next(iter(meta.owners))
+                    pool.driver_ctrl.send(
+                        owner, {"cmd": "xfer_send", "entry": eid, "dst": r}
+                    )
+                    pool.driver_ctrl.send(
+                        r, {"cmd": "xfer_recv", "entry": eid, "src": owner,
+                            "meta": meta.to_wire()}
+                    )
+                    wait_entries_per_rank[r].append(eid)
+                    meta.owners.add(r)
+                    METRICS.inc("lzy_transfers")
+                    METRICS.inc("lzy_transfer_bytes", meta.nbytes)
+
+        func_bytes = pickle_value(call.signature.func)
+        snap = self.workflow.snapshot
+        for i, r in enumerate(ranks):
+            spec = TaskSpec(
+                task_id=task_id,
+                name=call.callable_name,
+                func_bytes=func_bytes,
+                arg_entries=list(call.arg_entry_ids),
+                kwarg_entries=dict(call.kwarg_entry_ids),
+                output_entries=[
+                    (eid, snap.get_entry(eid).storage_uri) for eid in call.entry_ids
+                ],
+                exception_entry=call.exception_id,
+                env_vars=dict(call.env.env_variables),
+                cache=call.cache,
+                version=call.version,
+                storage_root=self.workflow.owner.storage_uri,
+                inline_values=specs_inline,
+                wait_entries=wait_entries_per_rank[r],
+                gang={**gang, "gang_rank": i} if gang is not None else None,
+            )
+            pool.driver_ctrl.send(r, {"cmd": "task", "spec": spec})
+            self.outstanding[r] += 1
+        self.inflight += 1
+        self.journal.record(task_id, "scheduled", call.callable_name)
+        METRICS.observe("lzy_dispatch", time.perf_counter() - dispatch_t0)
+
+    def _pick_rank(self, call: "LzyCall") -> int:
+        # data affinity: the rank already holding the most input bytes;
+        # ties broken by load
+        byrank: Dict[int, int] = {}
+        for eid in call.input_entry_ids():
+            meta = self.meta.get(eid)
+            if meta is None or meta.kind != KIND_TENSOR:
+                continue
+            for r in meta.owners:
+                byrank[r] = byrank.get(r, 0) + meta.nbytes
+        if byrank:
+            best = max(byrank.items(), key=lambda kv: (kv[1], -self.outstanding[kv[0]]))
+            return best[0]
+        return min(self.outstanding, key=lambda r: (self.outstanding[r], r))
+
+    def _pick_gang(self, k: int) -> List[int]:
+        ranks = sorted(
+            self.outstanding, key=lambda r: (self.outstanding[r], r)
+        )[:k]
+        return sorted(ranks)
+
+    # -- completion ----------------------------------------------------------
+
+    def _on_done(self, rank: int, result: TaskResult) -> bool:
+        """Returns True when the task fully completed (all gang members)."""
+        self.outstanding[rank] -= 1
+        call = self.calls[result.task_id]
+        gang = self.gang_pending.get(result.task_id)
+        primary = self.task_ranks[result.task_id][0]
+        if rank == primary:
+            self._record_outputs(call, rank, result)
+        if gang is not None:
+            gang.discard(rank)
+            if gang:
+                return False  # wait for the rest of the gang
+            self.gang_pending.pop(result.task_id, None)
+        self.inflight -= 1
+        self.journal.record(result.task_id, "done")
+        if result.cached:
+            METRICS.inc("lzy_cache_hits_pool")
+        return True
+
+    def _record_outputs(self, call: "LzyCall", rank: int, result: TaskResult) -> None:
+        snap = self.workflow.snapshot
+        for wire in result.outputs:
+            meta = EntryMeta.from_wire(wire)
+            meta.owners = {rank}
+            self.meta[meta.entry_id] = meta
+            if "uri" in wire:
+                snap.update_entry_uri(meta.entry_id, wire["uri"])
+
+    def _on_failed(self, rank: int, result: TaskResult) -> None:
+        self.outstanding[rank] -= 1
+        call = self.calls[result.task_id]
+        gang = self.gang_pending.get(result.task_id)
+        exc_name, exc_msg, tb = unpickle_value(result.exc_bytes)
+        self.workflow.snapshot.put(call.exception_id, (exc_name, exc_msg, tb))
+        self.errors.append(
+            LzyExecutionError(
+                f"Op {call.callable_name} failed on rank {rank}: "
+                f"{exc_name}: {exc_msg}",
+                task_id=result.task_id,
+                remote_traceback=tb,
+            )
+        )
+        self.journal.record(result.task_id, "failed", exc_msg)
+        if gang is not None:
+            gang.discard(rank)
+            if gang:
+                return  # remaining gang members still must report
+            self.gang_pending.pop(result.task_id, None)
+        self.inflight -= 1

@@ -1,0 +1,231 @@
+"""Picklable task description + worker-side execution.
+
+Re-design of the reference's worker workload: where the reference embeds a
+pickled ProcessingRequest in a command line and re-reads inputs from slot
+files (reference: pylzy api/v1/remote/runtime.py:368-384 +
+api/v1/startup.py:109-185), the pool sends a compact TaskSpec over the
+control plane and the worker reads inputs straight from its in-process
+store (already landed there by RCCL/xGMI transfers).
+"""
+from __future__ import annotations
+
+import logging
+import os
+import time
+import traceback
+from dataclasses import dataclass, field
+from typing import Any, Dict, List, Optional, Tuple
+
+from lzy_amd.runtime.executor import cache_base_uri
+from lzy_amd.utils.logs import OpLogCapture
+from lzy_amd.utils.metrics import METRICS
+
+_LOG = logging.getLogger("lzy_amd.taskspec")
+
+
+@dataclass
+class TaskSpec:
+    task_id: str
+    name: str
+    func_bytes: bytes
+    arg_entries: List[str]
+    kwarg_entries: Dict[str, str]
+    output_entries: List[Tuple[str, str]]  # (entry_id, storage_uri)
+    exception_entry: str
+    env_vars: Dict[str, str] = field(default_factory=dict)
+    cache: bool = False
+    version: str = "0.0"
+    storage_root: str = ""
+    inline_values: Dict[str, bytes] = field(default_factory=dict)
+    wait_entries: List[str] = field(default_factory=list)
+    gang: Optional[dict] = None  # {"ranks": [...], "gang_rank": i, "tag": str}
+
+
+@dataclass
+class TaskResult:
+    task_id: str
+    ok: bool
+    outputs: List[dict] = field(default_factory=list)  # EntryMeta wire dicts
+    exc_bytes: Optional[bytes] = None
+    elapsed_s: float = 0.0
+    cached: bool = False
+
+
+class WorkerStore:
+    """Per-rank value store: entry_id -> live python object (device tensors
+    stay in HBM).  Non-tensor values keep a pre-pickled byte image so
+    transfer sizes are known up front (single-phase recv)."""
+
+    def __init__(self) -> None:
+        self.values: Dict[str, Any] = {}
+        self.pickled: Dict[str, bytes] = {}
+
+    def put(self, entry_id: str, value: Any, pickled: Optional[bytes] = None) -> None:
+        self.values[entry_id] = value
+        if pickled is not None:
+            self.pickled[entry_id] = pickled
+
+    def get(self, entry_id: str) -> Any:
+        return self.values[entry_id]
+
+    def has(self, entry_id: str) -> bool:
+        return entry_id in self.values
+
+    def drop(self, entry_id: str) -> None:
+        self.values.pop(entry_id, None)
+        self.pickled.pop(entry_id, None)
+
+    def clear(self) -> None:
+        self.values.clear()
+        self.pickled.clear()
+
+
+def run_taskspec(
+    spec: TaskSpec,
+    store: WorkerStore,
+    serializers,
+    storage,
+    gang_group=None,
+) -> TaskResult:
+    """Execute one TaskSpec against the worker store (worker innermost loop)."""
+    from lzy_amd.channels.transport import describe_value, pickle_value, unpickle_value
+    from lzy_amd.snapshot import hash_value
+
+    import torch
+
+    t0 = time.perf_counter()
+
+    for eid, data in spec.inline_values.items():
+        if not store.has(eid):
+            store.put(eid, unpickle_value(data), pickled=data)
+
+    # -- result cache (worker-side CheckCache; rank-agnostic key) -----------
+    out_uris = dict(spec.output_entries)
+    if spec.cache:
+        hashes = [
+            hash_value(store.get(eid), serializers)
+            for eid in list(spec.arg_entries) + list(spec.kwarg_entries.values())
+        ]
+        base = cache_base_uri(spec.storage_root, spec.name, spec.version, hashes)
+        out_uris = {
+            eid: f"{base}/return_{i}"
+            for i, (eid, _) in enumerate(spec.output_entries)
+        }
+        if all(storage.blob_exists(u) for u in out_uris.values()):
+            outputs = []
+            for eid, uri in out_uris.items():
+                data = storage.read_bytes(uri)
+                fmt = _read_fmt(storage, uri)  # format recorded beside the blob
+                value = serializers.loads(data, fmt)
+                store.put(eid, value)
+                meta = describe_value(eid, value)
+                outputs.append({**meta.to_wire(), "uri": uri})
+            METRICS.inc("lzy_cache_hits", op=spec.name)
+            return TaskResult(
+                task_id=spec.task_id, ok=True, outputs=outputs,
+                elapsed_s=time.perf_counter() - t0, cached=True,
+            )
+
+    # -- materialize inputs --------------------------------------------------
+    try:
+        args = [store.get(eid) for eid in spec.arg_entries]
+        kwargs = {k: store.get(eid) for k, eid in spec.kwarg_entries.items()}
+    except KeyError as e:
+        return _fail(spec, t0, RuntimeError(f"missing input entry {e}"), "")
+
+    func = unpickle_value(spec.func_bytes)
+
+    capture = OpLogCapture.instance()
+    capture.route_current_thread(spec.name)
+    old_env: Dict[str, Optional[str]] = {}
+    gang_env: Dict[str, str] = {}
+    if spec.gang is not None:
+        gang_env = {
+            "LZY_OP_RANK": str(spec.gang["gang_rank"]),
+            "LZY_OP_WORLD_SIZE": str(len(spec.gang["ranks"])),
+        }
+    try:
+        for k, v in {**spec.env_vars, **gang_env}.items():
+            old_env[k] = os.environ.get(k)
+            os.environ[k] = v
+        if spec.gang is not None:
+            from lzy_amd.runtime.context import _set_op_context, OpContext
+
+            _set_op_context(OpContext(
+                gang_rank=spec.gang["gang_rank"],
+                gang_size=len(spec.gang["ranks"]),
+                ranks=tuple(spec.gang["ranks"]),
+                process_group=gang_group,
+            ))
+        result = func(*args, **kwargs)
+    except BaseException as e:  # noqa: BLE001 - transported as a value
+        return _fail(spec, t0, e, traceback.format_exc())
+    finally:
+        if spec.gang is not None:
+            from lzy_amd.runtime.context import _set_op_context
+
+            _set_op_context(None)
+        for k, v in old_env.items():
+            if v is None:
+                os.environ.pop(k, None)
+            else:
+                os.environ[k] = v
+        capture.unroute_current_thread()
+
+    # -- store outputs -------------------------------------------------------
+    n = len(spec.output_entries)
+    if n == 1:
+        outs: Tuple[Any, ...] = (result,)
+    elif isinstance(result, tuple) and len(result) == n:
+        outs = result
+    else:
+        return _fail(
+            spec, t0,
+            RuntimeError(
+                f"op {spec.name} declared {n} outputs, returned {type(result).__name__}"
+            ),
+            "",
+        )
+
+    outputs = []
+    gang_primary = spec.gang is None or spec.gang["gang_rank"] == 0
+    for (eid, _), value in zip(spec.output_entries, outs):
+        pickled = None
+        if not isinstance(value, torch.Tensor):
+            pickled = pickle_value(value)
+        store.put(eid, value, pickled=pickled)
+        meta = describe_value(eid, value)
+        if pickled is not None:
+            meta.nbytes = len(pickled)
+        outputs.append(meta.to_wire())
+        if spec.cache and gang_primary:
+            data, fmt = serializers.dumps(value)
+            storage.write_bytes(out_uris[eid], data)
+            _write_fmt(storage, out_uris[eid], fmt)
+
+    METRICS.inc("lzy_op_runs", op=spec.name)
+    elapsed = time.perf_counter() - t0
+    METRICS.observe("lzy_op_run", elapsed)
+    return TaskResult(task_id=spec.task_id, ok=True, outputs=outputs, elapsed_s=elapsed)
+
+
+def _fail(spec: TaskSpec, t0: float, exc: BaseException, tb: str) -> TaskResult:
+    from lzy_amd.channels.transport import pickle_value
+
+    METRICS.inc("lzy_op_failures", op=spec.name)
+    payload = pickle_value((type(exc).__name__, str(exc), tb))
+    return TaskResult(
+        task_id=spec.task_id, ok=False, exc_bytes=payload,
+        elapsed_s=time.perf_counter() - t0,
+    )
+
+
+def _write_fmt(storage, uri: str, fmt: str) -> None:
+    storage.write_bytes(uri + ".fmt", fmt.encode())
+
+
+def _read_fmt(storage, uri: str) -> str:
+    try:
+        return storage.read_bytes(uri + ".fmt").decode()
+    except Exception:
+        return "pickle"

@@ -94,6 +94,9 @@ class DefaultSnapshot:
         self._entries: Dict[str, SnapshotEntry] = {}
         self._values: Dict[str, Any] = {}
         self._hashes: Dict[str, str] = {}
+        # optional hook: called on a hot-tier miss to pull the value from a
+        # remote owner (GpuPoolRuntime wires this to an xGMI fetch)
+        self.fetcher = None
 
     @property
     def serializers(self) -> LzySerializerRegistry:
@@ -131,6 +134,10 @@ class DefaultSnapshot:
     def try_get(self, entry_id: str) -> TryGetResult:
         if entry_id in self._values:
             return TryGetResult(True, self._values[entry_id])
+        if self.fetcher is not None:
+            self.fetcher(entry_id)
+            if entry_id in self._values:
+                return TryGetResult(True, self._values[entry_id])
         entry = self._entries.get(entry_id)
         if entry is not None and self._storage.blob_exists(entry.storage_uri):
             value = self.load(entry_id)
@@ -155,19 +162,29 @@ class DefaultSnapshot:
     # -- durable tier -------------------------------------------------------
 
     def persist(self, entry_id: str) -> str:
-        """Serialize the hot value to its storage URI; returns the URI."""
+        """Serialize the hot value to its storage URI; returns the URI.
+
+        A ``.fmt`` sidecar records the actual data format so any process
+        (pool workers included) can load the blob without type context.
+        """
         entry = self._entries[entry_id]
         value = self._values[entry_id]
         data, fmt = self._serializers.dumps(value)
         entry.data_format = fmt
         self._storage.write_bytes(entry.storage_uri, data)
+        self._storage.write_bytes(entry.storage_uri + ".fmt", fmt.encode())
         self._hashes[entry_id] = _hash_bytes(data)
         return entry.storage_uri
 
     def load(self, entry_id: str) -> Any:
         entry = self._entries[entry_id]
         data = self._storage.read_bytes(entry.storage_uri)
-        value = self._serializers.loads(data, entry.data_format, entry.typ)
+        fmt = entry.data_format
+        try:
+            fmt = self._storage.read_bytes(entry.storage_uri + ".fmt").decode()
+        except Exception:
+            pass
+        value = self._serializers.loads(data, fmt, entry.typ)
         self._values[entry_id] = value
         return value
 

@@ -1,0 +1,58 @@
+"""Multi-process pool runtime tests: world_size 2 over gloo on CPU
+(the distributed-correctness harness the reference runs as in-thread
+multi-service contexts — SURVEY.md §4.3; here: real processes, real
+torch.distributed, no GPU required)."""
+import os
+import subprocess
+import sys
+from pathlib import Path
+
+import pytest
+
+ROOT = Path(__file__).resolve().parent.parent
+
+
+def _run_distributed(script: str, nproc: int, tmp_path, extra_env=None, timeout=180):
+    env = dict(os.environ)
+    env["LZY_AMD_STORAGE"] = str(tmp_path / "storage")
+    env["PYTHONPATH"] = str(ROOT) + os.pathsep + env.get("PYTHONPATH", "")
+    env["MASTER_ADDR"] = "127.0.0.1"
+    env.pop("RANK", None)
+    env.pop("WORLD_SIZE", None)
+    if extra_env:
+        env.update(extra_env)
+    cmd = [
+        sys.executable, "-m", "torch.distributed.run",
+        "--nnodes=1", f"--nproc-per-node={nproc}",
+        "--master-addr", "127.0.0.1", "--master-port", "0",
+        "--no-python" if False else script,
+    ]
+    return subprocess.run(
+        cmd, cwd=ROOT, env=env, capture_output=True, text=True, timeout=timeout
+    )
+
+
+def test_pool_two_ranks(tmp_path):
+    res = _run_distributed("tests/pool_script.py", 2, tmp_path)
+    if res.returncode != 0:
+        print("STDOUT:", res.stdout[-4000:])
+        print("STDERR:", res.stderr[-4000:])
+    assert res.returncode == 0
+    assert "POOL-SCRIPT-OK" in res.stdout
+
+
+def test_pool_single_rank_inprocess(tmp_path, monkeypatch):
+    """ws=1 degenerates to driver-only: full pipeline without dist."""
+    env = dict(os.environ)
+    env["LZY_AMD_STORAGE"] = str(tmp_path / "storage")
+    env.pop("RANK", None)
+    env.pop("WORLD_SIZE", None)
+    res = subprocess.run(
+        [sys.executable, "tests/pool_script_single.py"],
+        cwd=ROOT, env=env, capture_output=True, text=True, timeout=120,
+    )
+    if res.returncode != 0:
+        print("STDOUT:", res.stdout[-4000:])
+        print("STDERR:", res.stderr[-4000:])
+    assert res.returncode == 0
+    assert "SINGLE-OK" in res.stdout
