@@ -21,11 +21,16 @@ def _run_distributed(script: str, nproc: int, tmp_path, extra_env=None, timeout=
     env.pop("WORLD_SIZE", None)
     if extra_env:
         env.update(extra_env)
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
     cmd = [
         sys.executable, "-m", "torch.distributed.run",
         "--nnodes=1", f"--nproc-per-node={nproc}",
-        "--master-addr", "127.0.0.1", "--master-port", "0",
-        "--no-python" if False else script,
+        "--master-addr", "127.0.0.1", "--master-port", str(port),
+        script,
     ]
     return subprocess.run(
         cmd, cwd=ROOT, env=env, capture_output=True, text=True, timeout=timeout
@@ -45,6 +50,7 @@ def test_pool_single_rank_inprocess(tmp_path, monkeypatch):
     """ws=1 degenerates to driver-only: full pipeline without dist."""
     env = dict(os.environ)
     env["LZY_AMD_STORAGE"] = str(tmp_path / "storage")
+    env["PYTHONPATH"] = str(ROOT) + os.pathsep + env.get("PYTHONPATH", "")
     env.pop("RANK", None)
     env.pop("WORLD_SIZE", None)
     res = subprocess.run(
