@@ -66,6 +66,20 @@ _LOG = logging.getLogger("lzy_amd.pool")
 
 INLINE_LIMIT = 256 << 10  # CPU values up to 256 KiB travel inside the TaskSpec
 
+# function -> cloudpickle bytes memo (driver side; ops are module-level
+# callables so identity is stable and re-pickling per dispatch is waste)
+_FUNC_BYTES_CACHE: Dict[int, Tuple[Any, bytes]] = {}
+
+
+def _func_bytes(func) -> bytes:
+    key = id(func)
+    hit = _FUNC_BYTES_CACHE.get(key)
+    if hit is not None and hit[0] is func:
+        return hit[1]
+    data = pickle_value(func)
+    _FUNC_BYTES_CACHE[key] = (func, data)
+    return data
+
 
 def _storage_root() -> str:
     root = os.environ.get("LZY_AMD_STORAGE") or os.path.join(
@@ -428,12 +442,27 @@ class GpuPoolRuntime(Runtime):
         sched.run()
 
     def finish(self, workflow: "LzyWorkflow") -> None:
+        self._drop_workflow_entries(workflow)
         if self._journal is not None:
             self._journal.close()
             self._journal = None
 
     def abort(self, workflow: "LzyWorkflow") -> None:
         self.finish(workflow)
+
+    def _drop_workflow_entries(self, workflow: "LzyWorkflow") -> None:
+        """Free this workflow's values on every rank (HBM pressure control).
+
+        Persisted blobs (cache hits, whiteboard fields) survive on the
+        durable tier; post-exit proxy materialization falls back to them.
+        The reference behaves the same way structurally: VM death frees
+        slot memory, S3 keeps the persisted copies.
+        """
+        entry_ids = list(workflow.snapshot._entries.keys())
+        if not entry_ids:
+            return
+        pool = self.pool
+        pool.driver_ctrl.broadcast({"cmd": "drop_entries", "entries": entry_ids})
 
     # -- entry materialization on the driver --------------------------------
 
@@ -596,7 +625,7 @@ class _DriverScheduler:
                     METRICS.inc("lzy_transfers")
                     METRICS.inc("lzy_transfer_bytes", meta.nbytes)
 
-        func_bytes = pickle_value(call.signature.func)
+        func_bytes = _func_bytes(call.signature.func)
         snap = self.workflow.snapshot
         for i, r in enumerate(ranks):
             spec = TaskSpec(

@@ -133,7 +133,7 @@ def run_taskspec(
     except KeyError as e:
         return _fail(spec, t0, RuntimeError(f"missing input entry {e}"), "")
 
-    func = unpickle_value(spec.func_bytes)
+    func = _load_func(spec.func_bytes)
 
     capture = OpLogCapture.instance()
     capture.route_current_thread(spec.name)
@@ -207,6 +207,22 @@ def run_taskspec(
     elapsed = time.perf_counter() - t0
     METRICS.observe("lzy_op_run", elapsed)
     return TaskResult(task_id=spec.task_id, ok=True, outputs=outputs, elapsed_s=elapsed)
+
+
+_FUNC_CACHE: Dict[int, Any] = {}
+
+
+def _load_func(func_bytes: bytes):
+    """Worker-side unpickle memo keyed by content hash of the bytes."""
+    from lzy_amd.channels.transport import unpickle_value
+    from lzy_amd.sched import xxhash64
+
+    key = xxhash64(func_bytes)
+    func = _FUNC_CACHE.get(key)
+    if func is None:
+        func = unpickle_value(func_bytes)
+        _FUNC_CACHE[key] = func
+    return func
 
 
 def _fail(spec: TaskSpec, t0: float, exc: BaseException, tb: str) -> TaskResult:
