@@ -31,6 +31,18 @@ def _try_load() -> Optional[ctypes.CDLL]:
         _load_error = f"{_LIB_PATH} not built"
         return None
     try:
+        # Bind to torch's HIP runtime, not the system one: torch bundles
+        # its own libamdhip64.so.7 and two coexisting runtimes make our
+        # launches fail with hipErrorNoDevice.  Loading torch (and its
+        # runtime, RTLD_GLOBAL) first makes our DT_NEEDED resolve to the
+        # same runtime torch uses.
+        import torch  # noqa: F401
+
+        torch_hip = os.path.join(
+            os.path.dirname(torch.__file__), "lib", "libamdhip64.so"
+        )
+        if os.path.exists(torch_hip):
+            ctypes.CDLL(torch_hip, mode=ctypes.RTLD_GLOBAL)
         lib = ctypes.CDLL(_LIB_PATH)
     except OSError as e:  # e.g. no ROCm runtime on a CPU-only box
         _load_error = str(e)
