@@ -206,6 +206,7 @@ def run_taskspec(
     METRICS.inc("lzy_op_runs", op=spec.name)
     elapsed = time.perf_counter() - t0
     METRICS.observe("lzy_op_run", elapsed)
+    METRICS.observe(f"lzy_op::{spec.name}", elapsed)
     return TaskResult(task_id=spec.task_id, ok=True, outputs=outputs, elapsed_s=elapsed)
 
 
