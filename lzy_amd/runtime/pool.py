@@ -258,6 +258,7 @@ class GpuPool:
         self.agent: Optional[WorkerAgent] = None
         self.events: "queue.Queue[Tuple[int, dict]]" = queue.Queue()
         self._acks: Dict[str, Set[int]] = {}
+        self._ack_cv = threading.Condition()
         self._group_tags: Set[str] = set()
         self._seq = 0
 
@@ -322,20 +323,22 @@ class GpuPool:
     def _on_event(self, rank: int, msg: dict) -> None:
         ev = msg.get("ev")
         if ev in ("ack", "settled"):
-            self._acks.setdefault(f"{ev}:{msg['tag']}", set()).add(rank)
+            with self._ack_cv:
+                self._acks.setdefault(f"{ev}:{msg['tag']}", set()).add(rank)
+                self._ack_cv.notify_all()
         else:
             self.events.put((rank, msg))
 
     def wait_acks(self, kind: str, tag: str, ranks: Sequence[int], timeout: float = 300.0) -> None:
-        deadline = time.monotonic() + timeout
         key = f"{kind}:{tag}"
-        while time.monotonic() < deadline:
-            got = self._acks.get(key, set())
-            if set(ranks).issubset(got):
-                self._acks.pop(key, None)
-                return
-            time.sleep(0.0005)
-        raise TimeoutError(f"waiting for {kind} {tag} from {ranks}")
+        want = set(ranks)
+        with self._ack_cv:
+            ok = self._ack_cv.wait_for(
+                lambda: want.issubset(self._acks.get(key, set())), timeout
+            )
+            if not ok:
+                raise TimeoutError(f"waiting for {kind} {tag} from {ranks}")
+            self._acks.pop(key, None)
 
     def next_seq(self) -> int:
         self._seq += 1
@@ -413,6 +416,12 @@ class GpuPoolRuntime(Runtime):
         return StorageConfig(uri=f"file://{_storage_root()}")
 
     def start(self, workflow: "LzyWorkflow") -> None:
+        from lzy_amd.utils.metrics import timed
+
+        with timed("lzy_wf_start"):
+            self._start(workflow)
+
+    def _start(self, workflow: "LzyWorkflow") -> None:
         pool = self.pool  # workers never get past this line (serve loop)
         assert pool.is_driver
         # the rank-0 agent store backs the workflow snapshot: captured args
@@ -431,21 +440,30 @@ class GpuPoolRuntime(Runtime):
             os.path.join(self._journal_dir, f"{workflow.execution_id}.jsonl")
         )
         payload = workflow.owner.serializer_registry.user_serializers_payload()
-        tag = f"ser{pool.next_seq()}"
-        pool.driver_ctrl.broadcast(
-            {"cmd": "load_serializers", "payload": payload, "tag": tag}
-        )
-        pool.wait_acks("ack", tag, range(pool.world))
+        # ship user serializers only when they changed (usually: never)
+        if payload != getattr(pool, "_last_ser_payload", None):
+            tag = f"ser{pool.next_seq()}"
+            pool.driver_ctrl.broadcast(
+                {"cmd": "load_serializers", "payload": payload, "tag": tag}
+            )
+            pool.wait_acks("ack", tag, range(pool.world))
+            pool._last_ser_payload = payload
 
     def exec(self, workflow: "LzyWorkflow", calls: Sequence["LzyCall"]) -> None:
-        sched = _DriverScheduler(self.pool, workflow, calls, self._journal)
-        sched.run()
+        from lzy_amd.utils.metrics import timed
+
+        with timed("lzy_wf_exec"):
+            sched = _DriverScheduler(self.pool, workflow, calls, self._journal)
+            sched.run()
 
     def finish(self, workflow: "LzyWorkflow") -> None:
-        self._drop_workflow_entries(workflow)
-        if self._journal is not None:
-            self._journal.close()
-            self._journal = None
+        from lzy_amd.utils.metrics import timed
+
+        with timed("lzy_wf_finish"):
+            self._drop_workflow_entries(workflow)
+            if self._journal is not None:
+                self._journal.close()
+                self._journal = None
 
     def abort(self, workflow: "LzyWorkflow") -> None:
         self.finish(workflow)

@@ -306,7 +306,12 @@ static std::string json_escape(const std::string& s) {
 
 class Journal {
 public:
-    explicit Journal(const std::string& path) : path_(path) {
+    // sync=true fsyncs every record (strict crash-resume, e.g. fault tests);
+    // default is buffered-by-OS appends + fsync on close: losing a tail
+    // record on a crash only re-runs an idempotent task (the result cache
+    // dedups), and per-record fsync costs milliseconds on NVMe.
+    explicit Journal(const std::string& path, bool sync = false)
+        : path_(path), sync_(sync) {
         // mkdir -p parent
         std::string dir = path.substr(0, path.find_last_of('/'));
         if (!dir.empty()) {
@@ -332,11 +337,16 @@ public:
                            json_escape(detail) + "\"}\n";
         ssize_t n = ::write(fd_, line.data(), line.size());
         (void)n;
-        ::fsync(fd_);
+        if (sync_) ::fsync(fd_);
+    }
+
+    void sync() {
+        if (fd_ >= 0) ::fsync(fd_);
     }
 
     void close() {
         if (fd_ >= 0) {
+            ::fsync(fd_);
             ::close(fd_);
             fd_ = -1;
         }
@@ -344,6 +354,7 @@ public:
 
 private:
     std::string path_;
+    bool sync_ = false;
     int fd_ = -1;
 };
 
@@ -379,9 +390,11 @@ PYBIND11_MODULE(_core, m) {
         .def("__len__", &Dag::size);
 
     py::class_<Journal>(m, "Journal")
-        .def(py::init<const std::string&>(), py::arg("path"))
+        .def(py::init<const std::string&, bool>(), py::arg("path"),
+             py::arg("sync") = false)
         .def("record", &Journal::record, py::arg("tid"), py::arg("state"),
              py::arg("detail") = "")
+        .def("sync", &Journal::sync)
         .def("close", &Journal::close)
         .def_static("replay", [](const std::string& path) {
             // replay via python's json for robustness against torn tails

@@ -102,18 +102,26 @@ class PyJournal:
     """Append-only task-state journal for crash resume (reference analogue:
     the durable LRO step machine, long-running/OperationRunnerBase.java:27)."""
 
-    def __init__(self, path: str) -> None:
+    def __init__(self, path: str, sync: bool = False) -> None:
         self._path = path
+        self._sync = sync
         os.makedirs(os.path.dirname(path) or ".", exist_ok=True)
         self._f = open(path, "a", buffering=1)
 
     def record(self, tid: str, state: str, detail: str = "") -> None:
         self._f.write(json.dumps({"t": tid, "s": state, "d": detail}) + "\n")
         self._f.flush()
+        if self._sync:
+            os.fsync(self._f.fileno())
+
+    def sync(self) -> None:
         os.fsync(self._f.fileno())
 
     def close(self) -> None:
-        self._f.close()
+        if not self._f.closed:
+            self._f.flush()
+            os.fsync(self._f.fileno())
+            self._f.close()
 
     @staticmethod
     def replay(path: str) -> Dict[str, str]:
