@@ -225,7 +225,7 @@ def main() -> None:
         import sys as _sys
 
         for name, xs in sorted(METRICS._timings.items()):
-            if name.startswith(("lzy_op::", "lzy_wf_")) or name in ("lzy_dispatch", "lzy_graph_build"):
+            if name.startswith(("lzy_op::", "lzy_wf_")) or name in ("lzy_dispatch", "lzy_graph_build", "lzy_task_overhead"):
                 st = METRICS.timing_stats(name)
                 print(
                     f"# {name}: n={st['count']} mean={st['mean']*1e3:.2f}ms "

@@ -38,7 +38,7 @@ from typing import TYPE_CHECKING, Any, Dict, List, Optional, Sequence, Set, Tupl
 import torch
 import torch.distributed as dist
 
-from lzy_amd.channels.control import DriverControl, WorkerControl, broadcast_port
+from lzy_amd.channels.control import DriverControl, WorkerControl, broadcast_address
 from lzy_amd.channels.transport import (
     KIND_BYTES,
     KIND_TENSOR,
@@ -91,7 +91,7 @@ def _storage_root() -> str:
 class WorkerAgent:
     """Per-rank serve loop + executor thread (every rank, driver included)."""
 
-    def __init__(self, rank: int, world: int, port: int, pg_data, device):
+    def __init__(self, rank: int, world: int, address: str, pg_data, device):
         self.rank = rank
         self.world = world
         self.device = device
@@ -99,7 +99,7 @@ class WorkerAgent:
         self.serializers = LzySerializerRegistry()
         self.storage = FsStorageClient()
         self.transport = Transport(pg_data, device)
-        self.ctrl = WorkerControl(rank, port)
+        self.ctrl = WorkerControl(rank, address)
         self._pending: Dict[str, Tuple[list, Any]] = {}  # entry -> (works, fin)
         self._outbox: List[Tuple[list, Any]] = []
         self._groups: Dict[str, Any] = {}
@@ -288,17 +288,17 @@ class GpuPool:
 
         if self.is_driver:
             self.driver_ctrl = DriverControl(self.world, self._on_event)
-            port = self.driver_ctrl.port
+            address = self.driver_ctrl.address
             accept_thread = threading.Thread(
                 target=self.driver_ctrl.accept_all, daemon=True, name="lzy-accept"
             )
             accept_thread.start()
         else:
-            port = None
+            address = None
         if self.world > 1:
-            port = broadcast_port(port)
+            address = broadcast_address(address)
 
-        self.agent = WorkerAgent(self.rank, self.world, port, pg_data, self.device)
+        self.agent = WorkerAgent(self.rank, self.world, address, pg_data, self.device)
 
         if self.is_driver:
             accept_thread.join()
@@ -514,6 +514,7 @@ class _DriverScheduler:
         workflow._entry_meta = self.meta
         self.outstanding: Dict[int, int] = {r: 0 for r in range(pool.world)}
         self.task_ranks: Dict[str, List[int]] = {}
+        self.task_dispatch_ts: Dict[str, float] = {}
         self.gang_pending: Dict[str, Set[int]] = {}
         self.errors: List[BaseException] = []
         self.inflight = 0
@@ -667,6 +668,7 @@ class _DriverScheduler:
             pool.driver_ctrl.send(r, {"cmd": "task", "spec": spec})
             self.outstanding[r] += 1
         self.inflight += 1
+        self.task_dispatch_ts[task_id] = dispatch_t0
         self.journal.record(task_id, "scheduled", call.callable_name)
         METRICS.observe("lzy_dispatch", time.perf_counter() - dispatch_t0)
 
@@ -708,6 +710,13 @@ class _DriverScheduler:
             self.gang_pending.pop(result.task_id, None)
         self.inflight -= 1
         self.journal.record(result.task_id, "done")
+        ts = self.task_dispatch_ts.pop(result.task_id, None)
+        if ts is not None:
+            # full round trip minus the op itself = framework overhead/task
+            METRICS.observe(
+                "lzy_task_overhead",
+                max(0.0, time.perf_counter() - ts - result.elapsed_s),
+            )
         if result.cached:
             METRICS.inc("lzy_cache_hits_pool")
         return True
