@@ -95,11 +95,24 @@ def unpickle_value(data: bytes) -> Any:
 
 
 class Transport:
-    """Per-rank transfer engine over a dedicated data process group."""
+    """Per-rank transfer engine over a dedicated data process group.
 
-    def __init__(self, pg: Optional[dist.ProcessGroup], device: Optional[torch.device]):
+    ``cuda_p2p``: device tensors go directly over RCCL/xGMI.  When ranks
+    outnumber physical GPUs (test harnesses; oversubscribed pools) RCCL
+    cannot build a comm, so device tensors are staged through pinned host
+    memory and gloo — every rank computes the same predicate, so sender
+    and receiver always agree on the wire format.
+    """
+
+    def __init__(self, pg: Optional[dist.ProcessGroup], device: Optional[torch.device],
+                 world: int = 1):
         self._pg = pg
         self._device = device
+        self._cuda_p2p = (
+            device is not None
+            and torch.cuda.is_available()
+            and torch.cuda.device_count() >= world
+        )
 
     # -- send ---------------------------------------------------------------
 
@@ -109,6 +122,8 @@ class Transport:
             t = value.detach()
             if not t.is_contiguous():
                 t = t.contiguous()
+            if t.is_cuda and not self._cuda_p2p:
+                t = t.cpu()
             work = dist.isend(t, dst=dst, group=self._pg)
             return [work], t
         data = prepickled if prepickled is not None else pickle_value(value)
@@ -123,11 +138,16 @@ class Transport:
         finalize() -> the received value (call after works complete)."""
         if meta.kind == KIND_TENSOR:
             dtype = getattr(torch, meta.dtype)
-            if meta.device_type == "cuda" and self._device is not None:
+            want_cuda = meta.device_type == "cuda" and self._device is not None
+            if want_cuda and self._cuda_p2p:
                 buf = torch.empty(meta.shape, dtype=dtype, device=self._device)
-            else:
-                buf = torch.empty(meta.shape, dtype=dtype)
+                work = dist.irecv(buf, src=src, group=self._pg)
+                return [work], (lambda: buf)
+            buf = torch.empty(meta.shape, dtype=dtype)
             work = dist.irecv(buf, src=src, group=self._pg)
+            if want_cuda:
+                dev = self._device
+                return [work], (lambda: buf.to(dev, non_blocking=False))
             return [work], (lambda: buf)
         buf = torch.empty(meta.nbytes, dtype=torch.uint8)
         work = dist.irecv(buf, src=src, group=self._pg)

@@ -98,7 +98,7 @@ class WorkerAgent:
         self.store = WorkerStore()
         self.serializers = LzySerializerRegistry()
         self.storage = FsStorageClient()
-        self.transport = Transport(pg_data, device)
+        self.transport = Transport(pg_data, device, world=world)
         self.ctrl = WorkerControl(rank, address)
         self._pending: Dict[str, Tuple[list, Any]] = {}  # entry -> (works, fin)
         self._outbox: List[Tuple[list, Any]] = []
@@ -186,7 +186,9 @@ class WorkerAgent:
                     if self.device is not None:
                         torch.cuda.synchronize(self.device)
                     if dist.is_initialized():
-                        dist.barrier()
+                        # CPU all-reduce = gloo path: correct under GPU
+                        # oversubscription and free of RCCL rendezvous
+                        dist.all_reduce(torch.zeros(1))
                     if self.device is not None:
                         torch.cuda.synchronize(self.device)
                     self.ctrl.send_event(
@@ -275,6 +277,13 @@ class GpuPool:
             self.device = torch.device("cuda", local % torch.cuda.device_count())
             torch.cuda.set_device(self.device)
 
+        # one rank per physical GPU is the production shape; ranks may
+        # exceed GPUs only in test harnesses, where RCCL cannot build a
+        # comm and the data plane stages through gloo instead
+        cuda_ok = (
+            self.device is not None
+            and torch.cuda.device_count() >= self.world
+        )
         pg_data = None
         if self.world > 1:
             if not dist.is_initialized():
@@ -282,7 +291,7 @@ class GpuPool:
                     backend=None,
                     rank=self.rank,
                     world_size=self.world,
-                    device_id=self.device,
+                    device_id=self.device if cuda_ok else None,
                 )
             pg_data = dist.new_group()  # dedicated transfer group
 

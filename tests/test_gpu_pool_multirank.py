@@ -1,0 +1,42 @@
+"""2-rank pool on a single GPU box (oversubscribed): validates the whole
+distributed wiring — dist init, transfers of device tensors (staged via
+gloo when ranks > GPUs), gang ops, barriers — with real CUDA tensors.
+On an 8-GPU node the same flow runs with RCCL p2p over xGMI instead."""
+import os
+import socket
+import subprocess
+import sys
+from pathlib import Path
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+ROOT = Path(__file__).resolve().parent.parent
+
+
+@pytest.mark.skipif(not torch.cuda.is_available(), reason="no GPU")
+def test_pool_two_ranks_one_gpu(tmp_path):
+    env = dict(os.environ)
+    env["LZY_AMD_STORAGE"] = str(tmp_path / "storage")
+    env["PYTHONPATH"] = str(ROOT) + os.pathsep + env.get("PYTHONPATH", "")
+    env.pop("RANK", None)
+    env.pop("WORLD_SIZE", None)
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    res = subprocess.run(
+        [
+            sys.executable, "-m", "torch.distributed.run",
+            "--nnodes=1", "--nproc-per-node=2",
+            "--master-addr", "127.0.0.1", "--master-port", str(port),
+            "tests/pool_script_gpu.py",
+        ],
+        cwd=ROOT, env=env, capture_output=True, text=True, timeout=420,
+    )
+    if res.returncode != 0:
+        print("STDOUT:", res.stdout[-4000:])
+        print("STDERR:", res.stderr[-4000:])
+    assert res.returncode == 0
+    assert "POOL-GPU-OK" in res.stdout
