@@ -287,8 +287,11 @@ class GpuPool:
         pg_data = None
         if self.world > 1:
             if not dist.is_initialized():
+                # explicit composite: CPU collectives on gloo (control
+                # broadcasts, barriers), CUDA tensors on nccl (=RCCL)
+                backend = "cpu:gloo,cuda:nccl" if torch.cuda.is_available() else "gloo"
                 dist.init_process_group(
-                    backend=None,
+                    backend=backend,
                     rank=self.rank,
                     world_size=self.world,
                     device_id=self.device if cuda_ok else None,
