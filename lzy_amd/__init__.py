@@ -8,7 +8,7 @@ device-tensor channels over xGMI, HIP/CDNA4 kernels for the data-plane
 pack/cast/checksum ops, and RCCL (torch.distributed "nccl") for
 multi-GPU ops.
 """
-from lzy_amd.core.lzy import Lzy
+from lzy_amd.core.lzy import Lzy, lzy_auth
 from lzy_amd.core.op import op
 from lzy_amd.core.workflow import LzyWorkflow
 from lzy_amd.env.environment import LzyEnvironment
@@ -25,6 +25,7 @@ __version__ = "0.1.0"
 
 __all__ = [
     "Lzy",
+    "lzy_auth",
     "op",
     "LzyWorkflow",
     "LzyEnvironment",

@@ -24,6 +24,17 @@ from lzy_amd.whiteboards.index import WhiteboardIndexClient
 from lzy_amd.whiteboards.wb import ReadOnlyWhiteboard
 
 
+def lzy_auth(*, user: str, key_path: str = "", endpoint: str = "",
+             whiteboards_endpoint: str = "") -> None:
+    """Source-compatibility shim for the reference's cloud auth
+    (reference: pylzy/lzy/core/lzy.py:27-43).  The single-node runtime has
+    no remote service to authenticate against; the user name is recorded
+    for log/metadata attribution."""
+    os.environ["LZY_USER"] = user
+    if key_path:
+        os.environ["LZY_KEY_PATH"] = key_path
+
+
 def _default_runtime() -> Runtime:
     # under torchrun (one process per GPU) the pool runtime is the engine;
     # otherwise the in-process local runtime.
@@ -50,6 +61,12 @@ class Lzy(WithEnvironmentMixin):
                 self.storage_registry.register_storage(
                     "provided_default_storage", cfg, default=True
                 )
+
+    def auth(self, *, user: str, key_path: str = "", endpoint: str = "",
+             whiteboards_endpoint: str = "") -> "Lzy":
+        lzy_auth(user=user, key_path=key_path, endpoint=endpoint,
+                 whiteboards_endpoint=whiteboards_endpoint)
+        return self
 
     # -- storage ------------------------------------------------------------
 

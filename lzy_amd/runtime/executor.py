@@ -110,7 +110,7 @@ def execute_call(call: "LzyCall") -> None:
     }
 
     capture = OpLogCapture.instance()
-    capture.route_current_thread(name)
+    out_buf, err_buf = capture.route_current_thread(name)
     t0 = time.perf_counter()
     try:
         with _env_vars(call.env.env_variables):
@@ -128,6 +128,7 @@ def execute_call(call: "LzyCall") -> None:
         ) from e
     finally:
         capture.unroute_current_thread()
+        _archive_logs(call, out_buf.getvalue(), err_buf.getvalue())
 
     elapsed = time.perf_counter() - t0
     METRICS.observe("lzy_op_run", elapsed)
@@ -140,6 +141,29 @@ def execute_call(call: "LzyCall") -> None:
             snap.persist(eid)
 
     FAULTS.hit("executor.after_run")
+
+
+def _archive_logs(call: "LzyCall", out: str, err: str) -> None:
+    """Persist captured op std-logs beside the run's storage prefix.
+
+    Reference capability: worker stdout/stderr flow to a Kafka topic and
+    are archived to S3 by s3-sink (reference: lzy/s3-sink Job.java:38-86);
+    here the archive is one write to the durable tier per op that logged.
+    """
+    if not out and not err:
+        return
+    try:
+        wf = call.workflow
+        base = (
+            f"{wf.owner.storage_uri}/lzy_logs/{wf.execution_id}"
+            f"/{call.callable_name}-{call.id[:8]}"
+        )
+        if out:
+            wf.snapshot.storage.write_bytes(base + ".out", out.encode())
+        if err:
+            wf.snapshot.storage.write_bytes(base + ".err", err.encode())
+    except Exception:  # noqa: BLE001 - archiving must never fail the op
+        _LOG.warning("failed to archive logs for %s", call.callable_name)
 
 
 def _input_value(call: "LzyCall", entry_id: str, typ: type) -> Any:

@@ -670,6 +670,7 @@ class _DriverScheduler:
                 ],
                 exception_entry=call.exception_id,
                 env_vars=dict(call.env.env_variables),
+                execution_id=self.workflow.execution_id,
                 cache=call.cache,
                 version=call.version,
                 storage_root=self.workflow.owner.storage_uri,

@@ -33,6 +33,7 @@ class TaskSpec:
     output_entries: List[Tuple[str, str]]  # (entry_id, storage_uri)
     exception_entry: str
     env_vars: Dict[str, str] = field(default_factory=dict)
+    execution_id: str = ""
     cache: bool = False
     version: str = "0.0"
     storage_root: str = ""
@@ -136,7 +137,7 @@ def run_taskspec(
     func = _load_func(spec.func_bytes)
 
     capture = OpLogCapture.instance()
-    capture.route_current_thread(spec.name)
+    out_buf, err_buf = capture.route_current_thread(spec.name)
     old_env: Dict[str, Optional[str]] = {}
     gang_env: Dict[str, str] = {}
     if spec.gang is not None:
@@ -171,6 +172,7 @@ def run_taskspec(
             else:
                 os.environ[k] = v
         capture.unroute_current_thread()
+        _archive_logs(spec, storage, out_buf.getvalue(), err_buf.getvalue())
 
     # -- store outputs -------------------------------------------------------
     n = len(spec.output_entries)
@@ -208,6 +210,23 @@ def run_taskspec(
     METRICS.observe("lzy_op_run", elapsed)
     METRICS.observe(f"lzy_op::{spec.name}", elapsed)
     return TaskResult(task_id=spec.task_id, ok=True, outputs=outputs, elapsed_s=elapsed)
+
+
+def _archive_logs(spec: TaskSpec, storage, out: str, err: str) -> None:
+    """Worker-side std-log archive (reference s3-sink analogue)."""
+    if (not out and not err) or not spec.storage_root:
+        return
+    try:
+        base = (
+            f"{spec.storage_root}/lzy_logs/{spec.execution_id}"
+            f"/{spec.name}-{spec.task_id[:8]}"
+        )
+        if out:
+            storage.write_bytes(base + ".out", out.encode())
+        if err:
+            storage.write_bytes(base + ".err", err.encode())
+    except Exception:  # noqa: BLE001
+        _LOG.warning("failed to archive logs for %s", spec.name)
 
 
 _FUNC_CACHE: Dict[int, Any] = {}

@@ -1,0 +1,47 @@
+"""Storage garbage collection.
+
+Reference capability: GC of stale executions and expired VMs
+(lzy-service gc/GarbageCollector.java:21, allocator gc/*): here the
+collectible state is the durable tier — per-run snapshots and archived
+logs.  Cache blobs and whiteboards are retained by default (they are the
+long-lived artifacts); pass flags to collect them too.
+"""
+from __future__ import annotations
+
+import shutil
+import time
+from pathlib import Path
+from typing import Dict
+
+
+def gc_storage(
+    storage_root: str,
+    ttl_seconds: float = 7 * 24 * 3600,
+    collect_cache: bool = False,
+    collect_whiteboards: bool = False,
+    now: float | None = None,
+) -> Dict[str, int]:
+    """Remove expired run data under ``storage_root``; returns counts."""
+    now = now if now is not None else time.time()
+    root = Path(storage_root)
+    removed = {"runs": 0, "logs": 0, "cache": 0, "whiteboards": 0}
+
+    def _sweep(subdir: str, key: str) -> None:
+        base = root / subdir
+        if not base.is_dir():
+            return
+        for child in base.iterdir():
+            try:
+                if now - child.stat().st_mtime > ttl_seconds:
+                    shutil.rmtree(child, ignore_errors=True)
+                    removed[key] += 1
+            except FileNotFoundError:
+                continue
+
+    _sweep("lzy_runs", "runs")
+    _sweep("lzy_logs", "logs")
+    if collect_cache:
+        _sweep("lzy_cache", "cache")
+    if collect_whiteboards:
+        _sweep("whiteboards", "whiteboards")
+    return removed

@@ -147,3 +147,73 @@ def test_file_passing(lzy):
         f = make_file("payload")
         s = read_file(f)
         assert str(s) == "payload"
+
+
+def test_log_archive(lzy, storage_root):
+    @op
+    def chatty(x: int) -> int:
+        print("archived line")
+        return x
+
+    with lzy.workflow("logwf") as wf:
+        int(chatty(1))
+        exec_id = wf.execution_id
+    import glob
+
+    logs = glob.glob(str(storage_root / "lzy_logs" / exec_id / "*.out"))
+    assert len(logs) == 1
+    assert "archived line" in open(logs[0]).read()
+
+
+def test_storage_gc(tmp_path):
+    import os
+    import time
+
+    from lzy_amd.storage.gc import gc_storage
+
+    root = tmp_path / "s"
+    for sub in ["lzy_runs/old", "lzy_runs/new", "lzy_logs/old", "lzy_cache/old"]:
+        d = root / sub
+        d.mkdir(parents=True)
+        (d / "blob").write_text("x")
+    old = time.time() - 10_000
+    os.utime(root / "lzy_runs/old", (old, old))
+    os.utime(root / "lzy_logs/old", (old, old))
+    os.utime(root / "lzy_cache/old", (old, old))
+
+    removed = gc_storage(str(root), ttl_seconds=3600)
+    assert removed == {"runs": 1, "logs": 1, "cache": 0, "whiteboards": 0}
+    assert (root / "lzy_runs/new").exists()
+    assert not (root / "lzy_runs/old").exists()
+    assert (root / "lzy_cache/old").exists()
+
+    removed = gc_storage(str(root), ttl_seconds=3600, collect_cache=True)
+    assert removed["cache"] == 1
+
+
+def test_lzy_auth_shim(monkeypatch):
+    import lzy_amd
+
+    lzy_amd.lzy_auth(user="alice", key_path="/tmp/k")
+    assert os.environ["LZY_USER"] == "alice"
+
+
+def test_nested_workflow(lzy, storage_root):
+    from lzy_amd import Lzy
+    from lzy_amd.runtime.local import LocalRuntime
+
+    @op
+    def outer(x: int) -> int:
+        inner_lzy = Lzy(runtime=LocalRuntime())
+
+        @op
+        def inner(y: int) -> int:
+            return y * 10
+
+        with inner_lzy.workflow("inner-wf"):
+            r = inner(x)
+            return int(r) + 1
+
+    with lzy.workflow("outer-wf"):
+        v = outer(4)
+        assert int(v) == 41
