@@ -80,8 +80,17 @@ class WritableWhiteboard:
         fields: Dict[str, dataclasses.Field] = {}
         field_entries: Dict[str, str] = {}
         registry = owner.serializer_registry
+        # resolve string annotations (PEP 563 `from __future__ import
+        # annotations` makes f.type a string)
+        try:
+            import typing
+
+            hints = typing.get_type_hints(typ)
+        except Exception:
+            hints = {}
         for f in dataclasses.fields(typ):
-            ftyp = f.type if isinstance(f.type, type) else object
+            declared = hints.get(f.name, f.type)
+            ftyp = declared if isinstance(declared, type) else object
             ser = registry.find_serializer_by_type(ftyp)
             if ser is None:
                 raise TypeError(f"Cannot find serializer for whiteboard field {f.name}")
