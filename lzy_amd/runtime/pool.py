@@ -137,6 +137,22 @@ class WorkerAgent:
             meta = EntryMeta.from_wire(msg["meta"])
             works, fin = self.transport.irecv_value(meta, msg["src"])
             self._pending[msg["entry"]] = (works, fin)
+        elif cmd == "ipc_export":
+            from lzy_amd.channels.transport import export_ipc
+
+            data = export_ipc(self.store.get(msg["entry"]))
+            self.ctrl.send_event(
+                {"ev": "ack", "tag": msg["tag"], "rank": self.rank,
+                 "payload": data}
+            )
+        elif cmd == "ipc_import":
+            from lzy_amd.channels.transport import import_ipc
+
+            data = msg["data"]
+            dev = self.device
+            self._pending[msg["entry"]] = (
+                None, (lambda d=data: import_ipc(d, dev))
+            )
         elif cmd == "settle":
             self._exec_q.put(msg)
         elif cmd == "new_group":
@@ -217,8 +233,9 @@ class WorkerAgent:
             if pending is None:
                 continue
             works, fin = pending
-            for w in works:
-                w.wait()
+            if works:
+                for w in works:
+                    w.wait()
             value = fin()
             pickled = None
             if not isinstance(value, torch.Tensor):
@@ -260,6 +277,7 @@ class GpuPool:
         self.agent: Optional[WorkerAgent] = None
         self.events: "queue.Queue[Tuple[int, dict]]" = queue.Queue()
         self._acks: Dict[str, Set[int]] = {}
+        self._ack_payloads: Dict[str, Dict[int, Any]] = {}
         self._ack_cv = threading.Condition()
         self._group_tags: Set[str] = set()
         self._seq = 0
@@ -336,7 +354,10 @@ class GpuPool:
         ev = msg.get("ev")
         if ev in ("ack", "settled"):
             with self._ack_cv:
-                self._acks.setdefault(f"{ev}:{msg['tag']}", set()).add(rank)
+                key = f"{ev}:{msg['tag']}"
+                self._acks.setdefault(key, set()).add(rank)
+                if "payload" in msg:
+                    self._ack_payloads.setdefault(key, {})[rank] = msg["payload"]
                 self._ack_cv.notify_all()
         else:
             self.events.put((rank, msg))
@@ -351,6 +372,7 @@ class GpuPool:
             if not ok:
                 raise TimeoutError(f"waiting for {kind} {tag} from {ranks}")
             self._acks.pop(key, None)
+            return self._ack_payloads.pop(key, {})
 
     def next_seq(self) -> int:
         self._seq += 1
@@ -641,19 +663,44 @@ class _DriverScheduler:
                     eid
                 ) or pickle_value(self.pool.agent.store.get(eid))
                 continue
+            from lzy_amd.channels.transport import KIND_TENSOR as _KT, ipc_enabled
+
+            use_ipc = (
+                ipc_enabled()
+                and meta.kind == _KT
+                and meta.device_type == "cuda"
+            )
             for r in ranks:
                 if r not in meta.owners:
                     owner = 0 if 0 in meta.owners else next(iter(meta.owners))
-                    pool.driver_ctrl.send(
-                        owner, {"cmd": "xfer_send", "entry": eid, "dst": r}
-                    )
-                    pool.driver_ctrl.send(
-                        r, {"cmd": "xfer_recv", "entry": eid, "src": owner,
-                            "meta": meta.to_wire()}
-                    )
+                    if use_ipc:
+                        # zero-copy path: map the producer's HBM allocation
+                        # in the consumer (hipIpc; same GPU = no copy,
+                        # cross GPU = one xGMI DMA on import)
+                        if meta.ipc_handle is None:
+                            tag = f"ipc{pool.next_seq()}"
+                            pool.driver_ctrl.send(
+                                owner, {"cmd": "ipc_export", "entry": eid,
+                                        "tag": tag}
+                            )
+                            payloads = pool.wait_acks("ack", tag, [owner])
+                            meta.ipc_handle = payloads[owner]
+                        pool.driver_ctrl.send(
+                            r, {"cmd": "ipc_import", "entry": eid,
+                                "data": meta.ipc_handle}
+                        )
+                        METRICS.inc("lzy_transfers_ipc")
+                    else:
+                        pool.driver_ctrl.send(
+                            owner, {"cmd": "xfer_send", "entry": eid, "dst": r}
+                        )
+                        pool.driver_ctrl.send(
+                            r, {"cmd": "xfer_recv", "entry": eid, "src": owner,
+                                "meta": meta.to_wire()}
+                        )
+                        METRICS.inc("lzy_transfers")
                     wait_entries_per_rank[r].append(eid)
                     meta.owners.add(r)
-                    METRICS.inc("lzy_transfers")
                     METRICS.inc("lzy_transfer_bytes", meta.nbytes)
 
         func_bytes = _func_bytes(call.signature.func)

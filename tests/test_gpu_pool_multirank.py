@@ -17,8 +17,10 @@ ROOT = Path(__file__).resolve().parent.parent
 
 
 @pytest.mark.skipif(not torch.cuda.is_available(), reason="no GPU")
-def test_pool_two_ranks_one_gpu(tmp_path):
+@pytest.mark.parametrize("transport", ["rccl", "ipc"])
+def test_pool_two_ranks_one_gpu(tmp_path, transport):
     env = dict(os.environ)
+    env["LZY_CHANNEL_TRANSPORT"] = transport
     env["LZY_AMD_STORAGE"] = str(tmp_path / "storage")
     env["PYTHONPATH"] = str(ROOT) + os.pathsep + env.get("PYTHONPATH", "")
     env.pop("RANK", None)
