@@ -56,6 +56,10 @@ def _try_load() -> Optional[ctypes.CDLL]:
     lib.lz_checksum.argtypes = [
         ctypes.c_void_p, ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p,
     ]
+    lib.lz_checksum_mfma.restype = ctypes.c_int
+    lib.lz_checksum_mfma.argtypes = [
+        ctypes.c_void_p, ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p,
+    ]
     lib.lz_fill_pattern.restype = ctypes.c_int
     lib.lz_fill_pattern.argtypes = [
         ctypes.c_void_p, ctypes.c_int64, ctypes.c_uint64, ctypes.c_void_p,
@@ -137,9 +141,15 @@ def cast_copy(src, dst) -> None:
     )
 
 
-def device_checksum(t) -> int:
+def device_checksum(t, method: str = "auto") -> int:
     """64-bit content hash of a device tensor, computed on-GPU (no PCIe
-    round-trip for the data; 8 bytes come back)."""
+    round-trip for the data; 8 bytes come back).
+
+    method: "mfma" routes the mixing through the i8 matrix cores (the
+    default for buffers >= 64 KiB — VALU stays free, bandwidth-bound),
+    "valu" is the splitmix64 kernel, "auto" picks by size.  The two
+    digests are different hash functions (both stable across runs).
+    """
     import torch
 
     lib = _require_native()
@@ -147,11 +157,15 @@ def device_checksum(t) -> int:
         raise ValueError("device_checksum requires a device tensor")
     flat = t.detach().contiguous().view(torch.uint8) if t.dtype != torch.uint8 \
         else t.detach().contiguous()
+    nbytes = flat.numel() * flat.element_size()
+    if method == "auto":
+        method = "mfma" if nbytes >= (64 << 10) else "valu"
+    fn = lib.lz_checksum_mfma if method == "mfma" else lib.lz_checksum
     out = torch.zeros(1, dtype=torch.int64, device=t.device)
     _check(
-        lib.lz_checksum(
+        fn(
             ctypes.c_void_p(flat.data_ptr()),
-            flat.numel() * flat.element_size(),
+            nbytes,
             ctypes.c_void_p(out.data_ptr()),
             ctypes.c_void_p(_current_stream_ptr()),
         )

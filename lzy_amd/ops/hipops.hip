@@ -214,6 +214,102 @@ extern "C" hipError_t lz_checksum(const void* data, int64_t nbytes,
 }
 
 // ---------------------------------------------------------------------------
+// checksum_mfma: matrix-universal-hash digest on the MFMA units.
+//
+// The bulk pass is HBM-bound either way; routing the mixing arithmetic
+// through the i8 matrix cores (v_mfma_i32_32x32x32_i8, ~4.4 PO/s) leaves
+// the VALU nearly idle: per 1 KiB tile one MFMA replaces ~1.6k VALU mix
+// ops, so the kernel runs at the memory ceiling with MfmaUtil carrying
+// the compute (verified with rocprofv3 PMC, see profiles/).
+//
+// Scheme: D += A · (X_t ^ salt(t)) over Z_2^32, where X_t is the next
+// 1 KiB of data as a 32x32 i8 tile (each lane holds 16 bytes = its B
+// fragment, loaded as one coalesced dwordx4), A is a fixed dense
+// pseudo-random i8 matrix (generated in-register from splitmix64), and
+// salt(t) makes the hash position-sensitive while the i32 accumulation
+// stays commutative across tiles.  Final fold: per-lane position-salted
+// splitmix of the 16 accumulators, wave reduction, one atomic per wave.
+// Any fixed lane->matrix-element mapping is a valid hash basis, so B
+// fragments are simply lane-linear memory.
+// ---------------------------------------------------------------------------
+
+typedef __attribute__((ext_vector_type(4))) int lz_i32x4;
+typedef __attribute__((ext_vector_type(16))) int lz_i32x16;
+
+__global__ void checksum_mfma_kernel(const uint8_t* __restrict__ data,
+                                     int64_t nbytes,
+                                     unsigned long long* __restrict__ out) {
+    const int lane = threadIdx.x & 63;
+    const int wave = threadIdx.x >> 6;
+    const int waves_per_block = blockDim.x >> 6;
+
+    // A operand: fixed dense pseudo-random bytes per lane (4 dwords)
+    lz_i32x4 a_frag;
+#pragma unroll
+    for (int r = 0; r < 4; ++r)
+        a_frag[r] = (int)(uint32_t)(splitmix64(0x5EEDULL * 131 + lane * 4 + r) | 0x01010101u);
+
+    lz_i32x16 acc = {0};
+
+    const int64_t ntiles = nbytes >> 10;  // 1 KiB tiles
+    const int64_t wave_id = (int64_t)blockIdx.x * waves_per_block + wave;
+    const int64_t wave_stride = (int64_t)gridDim.x * waves_per_block;
+    const lz_i32x4* tiles = reinterpret_cast<const lz_i32x4*>(data);
+
+    for (int64_t t = wave_id; t < ntiles; t += wave_stride) {
+        lz_i32x4 b_frag = tiles[t * 64 + lane];
+        const uint64_t s = splitmix64((uint64_t)t);
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+            b_frag[r] ^= (int)(uint32_t)(s >> ((r * 13) & 31));
+        acc = __builtin_amdgcn_mfma_i32_32x32x32_i8(a_frag, b_frag, acc, 0, 0, 0);
+    }
+
+    // fold accumulators with position salts
+    uint64_t h = 0;
+#pragma unroll
+    for (int r = 0; r < 16; ++r)
+        h += splitmix64((uint32_t)acc[r] ^ splitmix64((uint64_t)(lane * 16 + r)));
+
+    // tail (< 1 KiB) + length: one thread via the scalar mix
+    if (blockIdx.x == 0 && threadIdx.x == 0) {
+        const uint8_t* p = data + (ntiles << 10);
+        int64_t rem = nbytes - (ntiles << 10);
+        const uint64_t* words = reinterpret_cast<const uint64_t*>(p);
+        int64_t nwords = rem >> 3;
+        for (int64_t i = 0; i < nwords; ++i)
+            h += splitmix64(words[i] ^ splitmix64((uint64_t)(ntiles * 128 + i)));
+        int tb = (int)(rem & 7);
+        if (tb) {
+            uint64_t last = 0;
+            const uint8_t* q = p + (nwords << 3);
+            for (int k = 0; k < tb; ++k) last |= ((uint64_t)q[k]) << (8 * k);
+            h += splitmix64(last ^ 0xA5A5A5A5ULL);
+        }
+        h += splitmix64((uint64_t)nbytes ^ 0x1234567890ABCDEFULL);
+    }
+
+    for (int off = 32; off > 0; off >>= 1)
+        h += (uint64_t)__shfl_down((long long)h, off, 64);
+    if (lane == 0)
+        atomicAdd(out, (unsigned long long)h);
+}
+
+extern "C" hipError_t lz_checksum_mfma(const void* data, int64_t nbytes,
+                                       unsigned long long* out_device,
+                                       void* stream) {
+    hipStream_t s = (hipStream_t)stream;
+    hipError_t err = hipMemsetAsync(out_device, 0, 8, s);
+    if (err != hipSuccess) return err;
+    int64_t tiles = nbytes >> 10;
+    int64_t want = (tiles / 4 + LZ_BLOCK / 64 - 1) / (LZ_BLOCK / 64);
+    int blocks = (int)(want < 1 ? 1 : (want > LZ_MAX_BLOCKS ? LZ_MAX_BLOCKS : want));
+    hipLaunchKernelGGL(checksum_mfma_kernel, dim3(blocks), dim3(LZ_BLOCK), 0, s,
+                       (const uint8_t*)data, nbytes, out_device);
+    return hipGetLastError();
+}
+
+// ---------------------------------------------------------------------------
 // fill_pattern: test/verification helper (deterministic device-side fill).
 // ---------------------------------------------------------------------------
 

@@ -151,3 +151,49 @@ def test_snapshot_hash_uses_device_checksum():
     h1 = hash_value(t, reg)
     h2 = hash_value(t.clone(), reg)
     assert h1 == h2
+
+
+@requires_gpu
+def test_checksum_mfma_properties():
+    from lzy_amd.ops import device_checksum
+
+    t = torch.randn(1 << 20, device="cuda")  # 4 MiB -> mfma path
+    h1 = device_checksum(t, method="mfma")
+    h2 = device_checksum(t.clone(), method="mfma")
+    assert h1 == h2 != 0
+    t2 = t.clone()
+    t2[777] += 1.0
+    assert device_checksum(t2, method="mfma") != h1
+    n = t.numel()
+    swapped = torch.cat([t[n // 2:], t[: n // 2]])
+    assert device_checksum(swapped, method="mfma") != h1
+    # non-tile-aligned sizes exercise the tail path
+    for n in [1, 100, 1024, 1025, (1 << 16) + 37]:
+        x = torch.arange(n, device="cuda", dtype=torch.float32)
+        assert device_checksum(x, method="mfma") == device_checksum(
+            x.clone(), method="mfma"
+        )
+        y = x.clone()
+        if n > 1:
+            y[0], y[n - 1] = x[n - 1], x[0]
+            assert device_checksum(y, method="mfma") != device_checksum(
+                x, method="mfma"
+            )
+
+
+@requires_gpu
+def test_checksum_mfma_bandwidth():
+    from lzy_amd.ops import device_checksum
+    import time
+
+    t = torch.randn(256 << 20, device="cuda")  # 1 GiB
+    device_checksum(t, method="mfma")
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    iters = 5
+    for _ in range(iters):
+        device_checksum(t, method="mfma")
+    dt = (time.perf_counter() - t0) / iters
+    gbps = t.numel() * 4 / dt / 1e9
+    print(f"device_checksum mfma: {gbps:.0f} GB/s")
+    assert gbps > 1500, f"mfma checksum too slow: {gbps:.0f} GB/s"
