@@ -71,6 +71,9 @@ class DriverControl:
         except (EOFError, OSError):
             if not self._closed:
                 _LOG.warning("control connection to rank %d closed", rank)
+                # liveness signal (reference analogue: allocator heartbeat
+                # miss -> dead-VM cleanup, AllocatorPrivateService)
+                self._on_event(rank, {"ev": "worker_lost"})
 
     def send(self, rank: int, msg: dict) -> None:
         with self._lock:
@@ -78,8 +81,11 @@ class DriverControl:
 
     def broadcast(self, msg: dict) -> None:
         with self._lock:
-            for conn in self._conns.values():
-                conn.send(msg)
+            for rank, conn in self._conns.items():
+                try:
+                    conn.send(msg)
+                except (OSError, BrokenPipeError):
+                    _LOG.warning("broadcast to dead rank %d skipped", rank)
 
     def close(self) -> None:
         self._closed = True

@@ -614,6 +614,26 @@ class _DriverScheduler:
                     LzyExecutionError(f"agent rank {rank}: {msg['error']}")
                 )
                 self.inflight -= 1
+            elif ev == "worker_lost":
+                # fail every task inflight on the dead rank; exclude the
+                # rank from further placement (reference: dead-VM cleanup)
+                self.outstanding.pop(rank, None)
+                dead = [
+                    tid for tid, rks in self.task_ranks.items()
+                    if rank in rks and self.task_dispatch_ts.get(tid) is not None
+                ]
+                for tid in dead:
+                    self.task_dispatch_ts.pop(tid, None)
+                    self.errors.append(LzyExecutionError(
+                        f"worker rank {rank} died while running "
+                        f"{self.calls[tid].callable_name}", task_id=tid,
+                    ))
+                    self.journal.record(tid, "failed", f"worker {rank} lost")
+                    if tid not in failed_tasks:
+                        failed_tasks.add(tid)
+                        for ct in dag.fail(tid):
+                            self.journal.record(ct, "cancelled")
+                    self.inflight -= 1
             # barrier_done etc. are routed via acks, not here
 
         if self.errors:

@@ -62,3 +62,34 @@ def test_pool_single_rank_inprocess(tmp_path, monkeypatch):
         print("STDERR:", res.stderr[-4000:])
     assert res.returncode == 0
     assert "SINGLE-OK" in res.stdout
+
+
+def test_worker_death_detected(tmp_path):
+    # launch ranks manually: torchrun would kill the driver the moment
+    # rank 1 exits, hiding exactly the recovery we want to observe
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    base_env = dict(os.environ)
+    base_env["LZY_AMD_STORAGE"] = str(tmp_path / "storage")
+    base_env["PYTHONPATH"] = str(ROOT) + os.pathsep + base_env.get("PYTHONPATH", "")
+    base_env.update(
+        MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port), WORLD_SIZE="2"
+    )
+    procs = []
+    for rank in (0, 1):
+        env = dict(base_env)
+        env["RANK"] = str(rank)
+        env["LOCAL_RANK"] = str(rank)
+        procs.append(subprocess.Popen(
+            [sys.executable, "tests/pool_script_death.py"],
+            cwd=ROOT, env=env, stdout=subprocess.PIPE, stderr=subprocess.PIPE,
+            text=True,
+        ))
+    out0, err0 = procs[0].communicate(timeout=120)
+    procs[1].wait(timeout=30)
+    assert procs[1].returncode == 7  # the injected hard exit
+    assert "DEATH-DETECTED" in out0, out0[-3000:] + err0[-2000:]
+    assert "DEATH-NOT-DETECTED" not in out0
