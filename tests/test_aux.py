@@ -217,3 +217,44 @@ def test_nested_workflow(lzy, storage_root):
     with lzy.workflow("outer-wf"):
         v = outer(4)
         assert int(v) == 41
+
+
+def test_env_shortcuts():
+    import warnings
+
+    from lzy_amd.env import shortcuts as sc
+
+    e = sc.gpu(4).combine(sc.cpu(8)).combine(sc.env_vars(A="1"))
+    assert e.provisioning.gpu_count == 4
+    assert e.provisioning.cpu_count == 8
+    assert e.env_variables == {"A": "1"}
+    with warnings.catch_warnings(record=True) as w:
+        warnings.simplefilter("always")
+        sc.docker_image("ubuntu:22.04")
+        assert any("ignored" in str(x.message) for x in w)
+    assert sc.auto_python().provisioning.gpu_count is None
+
+
+def test_status_endpoint(lzy, storage_root, tmp_path):
+    import json
+    import urllib.request
+
+    @op
+    def one(x: int) -> int:
+        return x
+
+    with lzy.workflow("statuswf"):
+        int(one(1))
+
+    from lzy_amd.storage.fs import uri_to_path
+    from lzy_amd.utils.status import serve_status
+
+    port = serve_status(storage_root=str(storage_root))
+    wfs = json.loads(
+        urllib.request.urlopen(f"http://127.0.0.1:{port}/workflows").read()
+    )
+    assert any(w["execution_id"].startswith("statuswf") for w in wfs)
+    wbs = json.loads(
+        urllib.request.urlopen(f"http://127.0.0.1:{port}/whiteboards").read()
+    )
+    assert isinstance(wbs, list)
