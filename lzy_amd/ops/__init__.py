@@ -66,6 +66,11 @@ def _try_load() -> Optional[ctypes.CDLL]:
     ]
     lib.lz_error_name.restype = ctypes.c_char_p
     lib.lz_error_name.argtypes = [ctypes.c_int]
+    lib.lz_set_max_blocks.restype = None
+    lib.lz_set_max_blocks.argtypes = [ctypes.c_int]
+    cap = os.environ.get("LZY_HIP_MAX_BLOCKS")
+    if cap:
+        lib.lz_set_max_blocks(int(cap))
     _lib = lib
     return lib
 

@@ -25,7 +25,10 @@
 #include <cstdint>
 
 #define LZ_BLOCK 256
-#define LZ_MAX_BLOCKS 2048  // 256 CUs x 8 blocks/CU
+#define LZ_MAX_BLOCKS 2048  // 256 CUs x 8 blocks/CU (default; runtime-tunable)
+
+static int g_max_blocks = LZ_MAX_BLOCKS;
+extern "C" void lz_set_max_blocks(int b) { g_max_blocks = b > 0 ? b : LZ_MAX_BLOCKS; }
 
 // ---------------------------------------------------------------------------
 // dtype codes shared with python (lzy_amd/ops/__init__.py)
@@ -107,7 +110,7 @@ template <typename SrcT>
 static hipError_t launch_cast_from(const void* src, void* dst, int dst_dtype,
                                    int64_t n, hipStream_t stream) {
     int64_t want = (n / 8 + LZ_BLOCK - 1) / LZ_BLOCK;
-    int blocks = (int)(want < 1 ? 1 : (want > LZ_MAX_BLOCKS ? LZ_MAX_BLOCKS : want));
+    int blocks = (int)(want < 1 ? 1 : (want > g_max_blocks ? g_max_blocks : want));
     switch (dst_dtype) {
 #define LZ_CASE(code, T)                                                      \
     case code:                                                                \
@@ -207,7 +210,7 @@ extern "C" hipError_t lz_checksum(const void* data, int64_t nbytes,
     hipError_t err = hipMemsetAsync(out_device, 0, 8, s);
     if (err != hipSuccess) return err;
     int64_t want = (nbytes / 16 + LZ_BLOCK - 1) / LZ_BLOCK;
-    int blocks = (int)(want < 1 ? 1 : (want > LZ_MAX_BLOCKS ? LZ_MAX_BLOCKS : want));
+    int blocks = (int)(want < 1 ? 1 : (want > g_max_blocks ? g_max_blocks : want));
     hipLaunchKernelGGL(checksum_kernel, dim3(blocks), dim3(LZ_BLOCK), 0, s,
                        (const uint8_t*)data, nbytes, out_device);
     return hipGetLastError();
@@ -256,7 +259,26 @@ __global__ void checksum_mfma_kernel(const uint8_t* __restrict__ data,
     const int64_t wave_stride = (int64_t)gridDim.x * waves_per_block;
     const lz_i32x4* tiles = reinterpret_cast<const lz_i32x4*>(data);
 
-    for (int64_t t = wave_id; t < ntiles; t += wave_stride) {
+    // 4 tiles per iteration: the loads issue back-to-back (4 KiB of HBM
+    // requests in flight per wave) before the dependent salt+MFMA chain —
+    // the kernel is HBM-latency-bound otherwise.
+    constexpr int U = 4;
+    int64_t t = wave_id;
+    for (; t + (U - 1) * wave_stride < ntiles; t += (int64_t)U * wave_stride) {
+        lz_i32x4 b[U];
+#pragma unroll
+        for (int u = 0; u < U; ++u)
+            b[u] = tiles[(t + u * wave_stride) * 64 + lane];
+#pragma unroll
+        for (int u = 0; u < U; ++u) {
+            const uint64_t s = splitmix64((uint64_t)(t + u * wave_stride));
+#pragma unroll
+            for (int r = 0; r < 4; ++r)
+                b[u][r] ^= (int)(uint32_t)(s >> ((r * 13) & 31));
+            acc = __builtin_amdgcn_mfma_i32_32x32x32_i8(a_frag, b[u], acc, 0, 0, 0);
+        }
+    }
+    for (; t < ntiles; t += wave_stride) {
         lz_i32x4 b_frag = tiles[t * 64 + lane];
         const uint64_t s = splitmix64((uint64_t)t);
 #pragma unroll
@@ -303,7 +325,7 @@ extern "C" hipError_t lz_checksum_mfma(const void* data, int64_t nbytes,
     if (err != hipSuccess) return err;
     int64_t tiles = nbytes >> 10;
     int64_t want = (tiles / 4 + LZ_BLOCK / 64 - 1) / (LZ_BLOCK / 64);
-    int blocks = (int)(want < 1 ? 1 : (want > LZ_MAX_BLOCKS ? LZ_MAX_BLOCKS : want));
+    int blocks = (int)(want < 1 ? 1 : (want > g_max_blocks ? g_max_blocks : want));
     hipLaunchKernelGGL(checksum_mfma_kernel, dim3(blocks), dim3(LZ_BLOCK), 0, s,
                        (const uint8_t*)data, nbytes, out_device);
     return hipGetLastError();
@@ -326,7 +348,7 @@ extern "C" hipError_t lz_fill_pattern(void* data, int64_t nbytes, uint64_t seed,
                                       void* stream) {
     int64_t nwords = nbytes >> 3;
     int64_t want = (nwords + LZ_BLOCK - 1) / LZ_BLOCK;
-    int blocks = (int)(want < 1 ? 1 : (want > LZ_MAX_BLOCKS ? LZ_MAX_BLOCKS : want));
+    int blocks = (int)(want < 1 ? 1 : (want > g_max_blocks ? g_max_blocks : want));
     hipLaunchKernelGGL(fill_pattern_kernel, dim3(blocks), dim3(LZ_BLOCK), 0,
                        (hipStream_t)stream, (uint64_t*)data, nwords, seed);
     return hipGetLastError();
