@@ -1,0 +1,130 @@
+"""Layered configuration (reference: Micronaut YAML + @ConfigurationProperties
+per service, e.g. lzy-service/config/LzyServiceConfig.java,
+allocator/configs/ServiceConfig.java; test overrides as CLI -key=value args
+in test-context/config/LzyConfig.java; Python side env vars in
+pylzy/lzy/api/v1/remote/lzy_service_client.py).
+
+Single-node equivalent: one typed config object for the whole runtime,
+resolved in precedence order
+
+    explicit overrides  >  environment (LZY_*)  >  YAML file  >  defaults
+
+The YAML file is looked up at ``$LZY_CONFIG`` then ``./lzy.yaml``.  Every
+field maps to an env var ``LZY_<FIELD>`` (upper-case).  ``Config.get()``
+returns the process-wide instance; tests may ``Config.reset(**overrides)``.
+"""
+from __future__ import annotations
+
+import os
+import threading
+from dataclasses import dataclass, field, fields
+from typing import Any, Dict, Optional
+
+
+def _coerce(value: str, typ: type) -> Any:
+    if typ is bool:
+        return value.strip().lower() in ("1", "true", "yes", "on")
+    if typ is int:
+        return int(value)
+    if typ is float:
+        return float(value)
+    return value
+
+
+@dataclass
+class Config:
+    """All runtime knobs in one place.
+
+    Fields mirror the reference's per-service configs that still make
+    sense on one node; the cloud-only ones (K8s credentials, YC disks,
+    Kafka brokers, S3 endpoints) have no equivalent here by design.
+    """
+
+    # storage (reference: storage vending, lzy-service GetOrCreateDefaultStorage)
+    storage: str = ""                  # root URI/path of the durable tier; "" -> tmp default
+    # channels (reference: channel-manager + slots transports)
+    channel_transport: str = "rccl"    # "rccl" | "ipc"
+    channel_chunk_mb: int = 256        # chunk size for large-tensor transfers
+    # HIP data-plane kernels
+    hip_max_blocks: int = 0            # 0 -> kernel default grid cap
+    # scheduler / pool
+    dispatch_workers: int = 0          # 0 -> auto (LocalRuntime thread pool size)
+    gang_timeout_s: float = 120.0      # gang-allocation wait bound
+    heartbeat_period_s: float = 2.0    # worker liveness probe period
+    # result cache / snapshot
+    cache_enabled: bool = True
+    # logs / metrics
+    log_archive: bool = True           # archive per-op std logs to durable tier
+    metrics_port: int = 0              # >0 -> serve /metrics on this port
+    status_port: int = 0               # >0 -> serve status endpoint on this port
+    # client-visible identity (reference: lzy_auth)
+    user: str = ""
+
+    _frozen_env: Dict[str, str] = field(default_factory=dict, repr=False)
+
+    # -- resolution ------------------------------------------------------
+    @staticmethod
+    def _load_yaml() -> Dict[str, Any]:
+        path = os.environ.get("LZY_CONFIG") or (
+            "lzy.yaml" if os.path.exists("lzy.yaml") else ""
+        )
+        if not path or not os.path.exists(path):
+            return {}
+        import yaml
+
+        with open(path) as f:
+            data = yaml.safe_load(f) or {}
+        if not isinstance(data, dict):
+            raise ValueError(f"config file {path} must hold a mapping")
+        return data
+
+    @classmethod
+    def resolve(cls, **overrides: Any) -> "Config":
+        data: Dict[str, Any] = {}
+        yaml_data = cls._load_yaml()
+        for f in fields(cls):
+            if f.name.startswith("_"):
+                continue
+            if f.name in yaml_data:
+                data[f.name] = yaml_data[f.name]
+            env_key = f"LZY_{f.name.upper()}"
+            if env_key in os.environ:
+                data[f.name] = _coerce(os.environ[env_key], f.type if isinstance(f.type, type) else type(f.default))  # type: ignore[arg-type]
+            if f.name in overrides and overrides[f.name] is not None:
+                data[f.name] = overrides[f.name]
+        # back-compat aliases kept from earlier revisions
+        if "storage" not in data and os.environ.get("LZY_AMD_STORAGE"):
+            data["storage"] = os.environ["LZY_AMD_STORAGE"]
+        if "channel_transport" not in data and os.environ.get("LZY_CHANNEL_TRANSPORT"):
+            data["channel_transport"] = os.environ["LZY_CHANNEL_TRANSPORT"]
+        if "hip_max_blocks" not in data and os.environ.get("LZY_HIP_MAX_BLOCKS"):
+            data["hip_max_blocks"] = int(os.environ["LZY_HIP_MAX_BLOCKS"])
+        cfg = cls(**data)
+        for f in fields(cls):
+            if not f.name.startswith("_"):
+                v = getattr(cfg, f.name)
+                want = f.type if isinstance(f.type, type) else type(f.default)
+                if want in (int, float, bool, str) and not isinstance(v, want):
+                    setattr(cfg, f.name, _coerce(str(v), want))
+        return cfg
+
+    # -- process-wide instance ------------------------------------------
+    _instance: "Optional[Config]" = None
+    _lock = threading.Lock()
+
+    @classmethod
+    def get(cls) -> "Config":
+        with cls._lock:
+            if cls._instance is None:
+                cls._instance = cls.resolve()
+            return cls._instance
+
+    @classmethod
+    def reset(cls, **overrides: Any) -> "Config":
+        with cls._lock:
+            cls._instance = cls.resolve(**overrides)
+            return cls._instance
+
+
+def get_config() -> Config:
+    return Config.get()

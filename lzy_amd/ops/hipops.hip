@@ -25,7 +25,11 @@
 #include <cstdint>
 
 #define LZ_BLOCK 256
-#define LZ_MAX_BLOCKS 2048  // 256 CUs x 8 blocks/CU (default; runtime-tunable)
+// Default grid cap: 4 workgroups per CU.  Swept on MI355X (profiles/
+// kernel_sweep.md): 1024 beats 2048..16384 on every data-plane kernel —
+// persistent grid-stride loops keep each XCD's L2 working set small and
+// the HBM streams long; larger grids only add launch/tail overhead.
+#define LZ_MAX_BLOCKS 1024  // 256 CUs x 4 blocks/CU (runtime-tunable)
 
 static int g_max_blocks = LZ_MAX_BLOCKS;
 extern "C" void lz_set_max_blocks(int b) { g_max_blocks = b > 0 ? b : LZ_MAX_BLOCKS; }
