@@ -129,7 +129,9 @@ def import_ipc(data: bytes, device: Optional[torch.device]) -> torch.Tensor:
 
 
 def ipc_enabled() -> bool:
-    return os.environ.get("LZY_CHANNEL_TRANSPORT", "rccl") == "ipc"
+    from lzy_amd.config import get_config
+
+    return get_config().channel_transport == "ipc"
 
 
 class Transport:

@@ -109,20 +109,38 @@ class Config:
         return cfg
 
     # -- process-wide instance ------------------------------------------
+    # The cache is keyed by a fingerprint of the LZY_* environment: a
+    # changed env (tests monkeypatching per test; lzy_auth) re-resolves,
+    # a stable env costs one dict scan per get().
     _instance: "Optional[Config]" = None
+    _fingerprint: Optional[tuple] = None
+    _pinned: bool = False
     _lock = threading.Lock()
+
+    @staticmethod
+    def _env_fingerprint() -> tuple:
+        return tuple(sorted(
+            (k, v) for k, v in os.environ.items() if k.startswith("LZY_")
+        ))
 
     @classmethod
     def get(cls) -> "Config":
         with cls._lock:
-            if cls._instance is None:
+            fp = cls._env_fingerprint()
+            if cls._instance is None or (not cls._pinned and fp != cls._fingerprint):
                 cls._instance = cls.resolve()
+                cls._fingerprint = fp
+                cls._pinned = False
             return cls._instance
 
     @classmethod
     def reset(cls, **overrides: Any) -> "Config":
+        """Re-resolve now; with overrides the result is pinned until the
+        next reset (explicit overrides outrank later env edits)."""
         with cls._lock:
             cls._instance = cls.resolve(**overrides)
+            cls._fingerprint = cls._env_fingerprint()
+            cls._pinned = bool(overrides)
             return cls._instance
 
 

@@ -40,9 +40,13 @@ DEFAULT_STORAGE_ENV = "LZY_AMD_STORAGE"
 
 
 def default_storage_uri() -> str:
-    root = os.environ.get(DEFAULT_STORAGE_ENV) or os.path.join(
+    from lzy_amd.config import get_config
+
+    root = get_config().storage or os.path.join(
         tempfile.gettempdir(), "lzy_amd_storage"
     )
+    if root.startswith("file://"):
+        return root
     return f"file://{root}"
 
 

@@ -82,10 +82,12 @@ def _func_bytes(func) -> bytes:
 
 
 def _storage_root() -> str:
-    root = os.environ.get("LZY_AMD_STORAGE") or os.path.join(
+    from lzy_amd.config import get_config
+
+    root = get_config().storage or os.path.join(
         tempfile.gettempdir(), "lzy_amd_storage"
     )
-    return root
+    return root[len("file://"):] if root.startswith("file://") else root
 
 
 class WorkerAgent:
