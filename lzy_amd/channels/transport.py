@@ -154,41 +154,116 @@ class Transport:
             and torch.cuda.device_count() >= world
         )
 
+    # -- chunking ------------------------------------------------------------
+    #
+    # Large values go over the wire as a sequence of fixed-size chunks
+    # (reference: slots' offset-resumable chunked `Read` streams,
+    # slots-api.proto:33-46, SlotInputTransfer.java:43-99).  Benefits here:
+    # the host-staged fallback never materializes more than one chunk's
+    # staging buffer ahead of the link, and a retried transfer can resume
+    # from the first incomplete chunk (`offset_chunks`) instead of
+    # resending the whole tensor.  Chunks of one transfer are issued
+    # back-to-back on the (src,dst) pair, so RCCL/gloo order-matching
+    # pairs them 1:1 with the receiver's chunk recvs.
+
+    def _chunk_elems(self, elem_size: int) -> int:
+        from lzy_amd.config import get_config
+
+        return max(1, (get_config().channel_chunk_mb << 20) // max(1, elem_size))
+
+    @staticmethod
+    def _chunks(flat: torch.Tensor, per: int, offset_chunks: int):
+        n = flat.numel()
+        starts = range(offset_chunks * per, n, per)
+        return [flat[s: min(s + per, n)] for s in starts]
+
     # -- send ---------------------------------------------------------------
 
-    def isend_value(self, value: Any, prepickled: Optional[bytes], dst: int):
-        """Issue non-blocking send(s); returns (works, keepalive)."""
+    def isend_value(self, value: Any, prepickled: Optional[bytes], dst: int,
+                    offset_chunks: int = 0):
+        """Issue non-blocking send(s); returns (works, keepalive).
+
+        ``offset_chunks`` resumes a partially completed transfer: the
+        first ``offset_chunks`` chunks are assumed delivered and skipped.
+        """
         if isinstance(value, torch.Tensor):
             t = value.detach()
             if not t.is_contiguous():
                 t = t.contiguous()
             if t.is_cuda and not self._cuda_p2p:
                 t = t.cpu()
-            work = dist.isend(t, dst=dst, group=self._pg)
-            return [work], t
+            per = self._chunk_elems(t.element_size())
+            if t.numel() <= per and offset_chunks == 0:
+                work = dist.isend(t, dst=dst, group=self._pg)
+                return [work], t
+            flat = t.view(-1)
+            works = [
+                dist.isend(c, dst=dst, group=self._pg)
+                for c in self._chunks(flat, per, offset_chunks)
+            ]
+            return works, t
         data = prepickled if prepickled is not None else pickle_value(value)
         buf = torch.frombuffer(bytearray(data), dtype=torch.uint8)
-        work = dist.isend(buf, dst=dst, group=self._pg)
-        return [work], buf
+        per = self._chunk_elems(1)
+        if buf.numel() <= per and offset_chunks == 0:
+            work = dist.isend(buf, dst=dst, group=self._pg)
+            return [work], buf
+        works = [
+            dist.isend(c, dst=dst, group=self._pg)
+            for c in self._chunks(buf, per, offset_chunks)
+        ]
+        return works, buf
 
     # -- recv ---------------------------------------------------------------
 
-    def irecv_value(self, meta: EntryMeta, src: int):
+    def irecv_value(self, meta: EntryMeta, src: int, offset_chunks: int = 0,
+                    into: Optional[torch.Tensor] = None):
         """Issue non-blocking recv; returns (works, finalize) where
-        finalize() -> the received value (call after works complete)."""
+        finalize() -> the received value (call after works complete).
+
+        ``offset_chunks``/``into`` resume into an existing buffer that
+        already holds the first ``offset_chunks`` chunks."""
         if meta.kind == KIND_TENSOR:
             dtype = getattr(torch, meta.dtype)
             want_cuda = meta.device_type == "cuda" and self._device is not None
-            if want_cuda and self._cuda_p2p:
-                buf = torch.empty(meta.shape, dtype=dtype, device=self._device)
-                work = dist.irecv(buf, src=src, group=self._pg)
-                return [work], (lambda: buf)
-            buf = torch.empty(meta.shape, dtype=dtype)
-            work = dist.irecv(buf, src=src, group=self._pg)
-            if want_cuda:
-                dev = self._device
-                return [work], (lambda: buf.to(dev, non_blocking=False))
-            return [work], (lambda: buf)
-        buf = torch.empty(meta.nbytes, dtype=torch.uint8)
-        work = dist.irecv(buf, src=src, group=self._pg)
-        return [work], (lambda: unpickle_value(buf.numpy().tobytes()))
+            on_device = want_cuda and self._cuda_p2p
+            dev = self._device if on_device else None
+            per = self._chunk_elems(torch.empty(0, dtype=dtype).element_size())
+            buf = into if into is not None else torch.empty(
+                meta.shape, dtype=dtype, device=dev
+            )
+            n = buf.numel()
+            if n <= per and offset_chunks == 0:
+                works = [dist.irecv(buf, src=src, group=self._pg)]
+            else:
+                flat = buf.view(-1)
+                works = [
+                    dist.irecv(c, src=src, group=self._pg)
+                    for c in self._chunks(flat, per, offset_chunks)
+                ]
+            if want_cuda and not on_device:
+                d = self._device
+                return works, (lambda: buf.to(d, non_blocking=False))
+            return works, (lambda: buf)
+        per = self._chunk_elems(1)
+        buf = into if into is not None else torch.empty(meta.nbytes, dtype=torch.uint8)
+        if buf.numel() <= per and offset_chunks == 0:
+            works = [dist.irecv(buf, src=src, group=self._pg)]
+        else:
+            works = [
+                dist.irecv(c, src=src, group=self._pg)
+                for c in self._chunks(buf, per, offset_chunks)
+            ]
+        return works, (lambda: unpickle_value(buf.numpy().tobytes()))
+
+    @staticmethod
+    def completed_chunks(works: list) -> int:
+        """Number of leading completed chunk transfers — the resume offset
+        for a retry (reference: slots Read offset resume)."""
+        done = 0
+        for w in works:
+            if w is not None and w.is_completed():
+                done += 1
+            else:
+                break
+        return done

@@ -93,3 +93,9 @@ def test_worker_death_detected(tmp_path):
     assert procs[1].returncode == 7  # the injected hard exit
     assert "DEATH-DETECTED" in out0, out0[-3000:] + err0[-2000:]
     assert "DEATH-NOT-DETECTED" not in out0
+
+
+def test_chunked_transport_offset_resume(tmp_path):
+    r = _run_distributed("tests/chunked_transport_script.py", 2, tmp_path)
+    assert r.returncode == 0, r.stdout[-3000:] + r.stderr[-3000:]
+    assert "CHUNKED-TRANSPORT-OK" in r.stdout
