@@ -1,0 +1,82 @@
+"""End-to-end scenario tier (reference: test/.../PyApiTest.java +
+pylzy/tests/scenarios/<name>/ — each scenario is a directory with a
+script and an ``expected_stdout`` asserted line-by-line,
+PythonContextTests.java:197-246).
+
+Here each scenario under tests/scenarios/<name>/ has ``__init__.py``
+(the script, run as ``python -m tests.scenarios.<name>`` in a fresh
+process with an isolated storage root) and ``expected_stdout``.
+Expected lines match exactly, except lines starting with ``~`` which are
+regex-matched, and ``...`` which skips any number of output lines until
+the next expected line matches.
+"""
+import os
+import re
+import subprocess
+import sys
+from pathlib import Path
+
+import pytest
+
+ROOT = Path(__file__).resolve().parent.parent
+SCEN = Path(__file__).resolve().parent / "scenarios"
+
+_names = sorted(
+    d.name for d in SCEN.iterdir()
+    if d.is_dir() and (d / "expected_stdout").exists()
+) if SCEN.exists() else []
+
+
+def _match(expected_lines, actual_lines, scenario):
+    ei, ai = 0, 0
+    while ei < len(expected_lines):
+        exp = expected_lines[ei]
+        if exp == "...":
+            ei += 1
+            if ei >= len(expected_lines):
+                return  # trailing ... matches the rest
+            nxt = expected_lines[ei]
+            while ai < len(actual_lines) and not _line_ok(nxt, actual_lines[ai]):
+                ai += 1
+            if ai >= len(actual_lines):
+                pytest.fail(
+                    f"[{scenario}] expected line {ei} not found after '...': "
+                    f"{nxt!r}\n--- actual ---\n" + "\n".join(actual_lines)
+                )
+            continue
+        if ai >= len(actual_lines) or not _line_ok(exp, actual_lines[ai]):
+            got = actual_lines[ai] if ai < len(actual_lines) else "<eof>"
+            pytest.fail(
+                f"[{scenario}] line {ai}: expected {exp!r}, got {got!r}\n"
+                "--- actual ---\n" + "\n".join(actual_lines)
+            )
+        ei += 1
+        ai += 1
+
+
+def _line_ok(exp: str, actual: str) -> bool:
+    if exp.startswith("~"):
+        return re.fullmatch(exp[1:], actual) is not None
+    return exp == actual
+
+
+@pytest.mark.parametrize("name", _names)
+def test_scenario(name, tmp_path):
+    env = dict(os.environ)
+    env["LZY_AMD_STORAGE"] = str(tmp_path / "storage")
+    env["PYTHONPATH"] = str(ROOT) + os.pathsep + env.get("PYTHONPATH", "")
+    env["LZY_SCENARIO_TMP"] = str(tmp_path)
+    r = subprocess.run(
+        [sys.executable, str(SCEN / name / "__init__.py")],
+        cwd=ROOT, env=env, capture_output=True, text=True, timeout=300,
+    )
+    assert r.returncode == 0, (
+        f"[{name}] rc={r.returncode}\n--- stdout ---\n{r.stdout[-4000:]}"
+        f"\n--- stderr ---\n{r.stderr[-4000:]}"
+    )
+    expected = (SCEN / name / "expected_stdout").read_text().splitlines()
+    # op std-logs reach the client prefixed "[LZY-<task>] " (reference:
+    # "[LZY-REMOTE-...]", runtime.py:283-301) — strip the prefix so
+    # expectations read like plain program output
+    actual = [re.sub(r"^\[LZY-[^\]]*\] ", "", l) for l in r.stdout.splitlines()]
+    _match(expected, actual, name)
