@@ -99,3 +99,11 @@ def test_chunked_transport_offset_resume(tmp_path):
     r = _run_distributed("tests/chunked_transport_script.py", 2, tmp_path)
     assert r.returncode == 0, r.stdout[-3000:] + r.stderr[-3000:]
     assert "CHUNKED-TRANSPORT-OK" in r.stdout
+
+
+def test_bench_dag_four_ranks(tmp_path):
+    """The driver's 8-GPU scale shape, at world 4 on gloo: fan-out,
+    cross-rank tree merge, scalar gather, repeated steps."""
+    r = _run_distributed("tests/bench_dag_script.py", 4, tmp_path, timeout=240)
+    assert r.returncode == 0, r.stdout[-3000:] + r.stderr[-3000:]
+    assert "BENCH-DAG-OK" in r.stdout
