@@ -258,3 +258,13 @@ def test_status_endpoint(lzy, storage_root, tmp_path):
         urllib.request.urlopen(f"http://127.0.0.1:{port}/whiteboards").read()
     )
     assert isinstance(wbs, list)
+
+
+def test_status_dashboard_html(storage_root):
+    import urllib.request
+
+    from lzy_amd.utils.status import serve_status
+
+    port = serve_status(storage_root=str(storage_root))
+    html = urllib.request.urlopen(f"http://127.0.0.1:{port}/").read().decode()
+    assert "<html" in html and "Whiteboards" in html and "/metrics" in html
