@@ -47,6 +47,7 @@ class Config:
     channel_chunk_mb: int = 256        # chunk size for large-tensor transfers
     # HIP data-plane kernels
     hip_max_blocks: int = 0            # 0 -> kernel default grid cap
+    op_streams: int = 4                # HIP streams per device for op overlap
     # scheduler / pool
     dispatch_workers: int = 0          # 0 -> auto (LocalRuntime thread pool size)
     gang_timeout_s: float = 120.0      # gang-allocation wait bound

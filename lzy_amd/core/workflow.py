@@ -94,6 +94,12 @@ class LzyWorkflow:
         finally:
             self._finished = True
             LzyWorkflow._active.wf = None
+            # values may outlive the workflow as plain tensors: guarantee
+            # their producing streams completed, then free the events
+            from lzy_amd.runtime.streams import STREAMS
+
+            for eid in list(self.snapshot._values):
+                STREAMS.sync_and_drop(eid)
         return False
 
     # -- calls & barrier ------------------------------------------------------

@@ -109,11 +109,27 @@ def execute_call(call: "LzyCall") -> None:
         for k, eid in call.kwarg_entry_ids.items()
     }
 
+    # per-op HIP stream: independent ops on one GPU overlap; inputs are
+    # made safe with device-side event waits (lzy_amd/runtime/streams.py)
+    from contextlib import nullcontext
+
+    import torch
+
+    from lzy_amd.runtime.streams import STREAMS
+
+    stream = STREAMS.next_stream()
+    stream_ctx = torch.cuda.stream(stream) if stream is not None else nullcontext()
+
     capture = OpLogCapture.instance()
     out_buf, err_buf = capture.route_current_thread(name)
     t0 = time.perf_counter()
     try:
-        with _env_vars(call.env.env_variables):
+        with _env_vars(call.env.env_variables), stream_ctx:
+            for eid, v in zip(
+                list(call.arg_entry_ids) + list(call.kwarg_entry_ids.values()),
+                args + list(kwargs.values()),
+            ):
+                STREAMS.wait_value(eid, v)
             result = call.signature.func(*args, **kwargs)
     except BaseException as e:  # noqa: BLE001 - transported as a value
         elapsed = time.perf_counter() - t0
@@ -134,7 +150,7 @@ def execute_call(call: "LzyCall") -> None:
     METRICS.observe("lzy_op_run", elapsed)
     METRICS.inc("lzy_op_runs", op=name)
 
-    _store_outputs(call, result)
+    _store_outputs(call, result, stream=stream)
 
     if call.cache:
         for eid in call.entry_ids:
@@ -180,7 +196,9 @@ def _input_value(call: "LzyCall", entry_id: str, typ: type) -> Any:
     return materialize(value) if is_lzy_proxy(value) else value
 
 
-def _store_outputs(call: "LzyCall", result: Any) -> None:
+def _store_outputs(call: "LzyCall", result: Any, stream=None) -> None:
+    from lzy_amd.runtime.streams import STREAMS
+
     snap = call.workflow.snapshot
     n = len(call.entry_ids)
     if n == 1:
@@ -195,4 +213,6 @@ def _store_outputs(call: "LzyCall", result: Any) -> None:
             )
         outputs = result
     for eid, value in zip(call.entry_ids, outputs):
-        snap.put(eid, materialize(value) if is_lzy_proxy(value) else value)
+        value = materialize(value) if is_lzy_proxy(value) else value
+        snap.put(eid, value)
+        STREAMS.record_output(eid, value, stream=stream)

@@ -133,7 +133,13 @@ class DefaultSnapshot:
 
     def try_get(self, entry_id: str) -> TryGetResult:
         if entry_id in self._values:
-            return TryGetResult(True, self._values[entry_id])
+            value = self._values[entry_id]
+            # cross-stream hand-off safety: the reader's stream waits on
+            # the producing op's completion event (runtime/streams.py)
+            from lzy_amd.runtime.streams import STREAMS
+
+            STREAMS.wait_value(entry_id, value)
+            return TryGetResult(True, value)
         if self.fetcher is not None:
             self.fetcher(entry_id)
             if entry_id in self._values:
@@ -158,6 +164,9 @@ class DefaultSnapshot:
 
     def drop_value(self, entry_id: str) -> None:
         self._values.pop(entry_id, None)
+        from lzy_amd.runtime.streams import STREAMS
+
+        STREAMS.drop(entry_id)
 
     # -- durable tier -------------------------------------------------------
 
@@ -169,6 +178,9 @@ class DefaultSnapshot:
         """
         entry = self._entries[entry_id]
         value = self._values[entry_id]
+        from lzy_amd.runtime.streams import STREAMS
+
+        STREAMS.wait_value(entry_id, value)  # D2H copy orders after producer
         data, fmt = self._serializers.dumps(value)
         entry.data_format = fmt
         self._storage.write_bytes(entry.storage_uri, data)

@@ -27,6 +27,10 @@ class Metrics:
         with self._lock:
             self._counters[(name, tuple(sorted(labels.items())))] += value
 
+    def counter_value(self, name: str, **labels: str) -> float:
+        with self._lock:
+            return self._counters.get((name, tuple(sorted(labels.items()))), 0.0)
+
     def set_gauge(self, name: str, value: float, **labels: str) -> None:
         with self._lock:
             self._gauges[(name, tuple(sorted(labels.items())))] = value

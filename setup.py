@@ -62,12 +62,23 @@ class BuildExt(build_ext):
         subprocess.check_call(cmd)
 
 
+# Sanitizer builds (the JVM reference needs none; a C++
+# runtime does): LZY_SANITIZE=thread|address instruments the C++ core.
+# Run the suite with the matching libtsan/libasan preloaded, e.g.
+#   LZY_SANITIZE=thread python setup.py build_ext --inplace --force
+#   GLIBC_TUNABLES=glibc.rtld.optional_static_tls=4194304 \
+#     LD_PRELOAD=$(gcc -print-file-name=libtsan.so) pytest tests -m "not gpu"
+# (the TLS tunable avoids "cannot allocate memory in static TLS block")
+_SAN = os.environ.get("LZY_SANITIZE", "")
+_san_args = [f"-fsanitize={_SAN}", "-g", "-fno-omit-frame-pointer"] if _SAN else []
+
 core_ext = Extension(
     "lzy_amd.sched._core",
     sources=["lzy_amd/sched/core.cpp"],
     include_dirs=[_pybind11_include()],
     language="c++",
-    extra_compile_args=["-O2", "-std=c++17", "-fvisibility=hidden"],
+    extra_compile_args=["-O2", "-std=c++17", "-fvisibility=hidden"] + _san_args,
+    extra_link_args=_san_args,
 )
 
 setup(

@@ -1,0 +1,39 @@
+"""Stream placement / race-guard tests (CPU: everything must no-op;
+GPU behavior is covered in tests/test_gpu_workflow.py)."""
+import torch
+
+from lzy_amd.runtime.streams import STREAMS, StreamPlacer
+
+
+def test_cpu_noop():
+    p = StreamPlacer()
+    if not torch.cuda.is_available():
+        assert p.enabled is False
+        assert p.next_stream() is None
+    t = torch.ones(4)
+    p.record_output("e1", t)
+    p.wait_value("e1", t)      # cpu tensor: ignored
+    p.wait_value("missing", t)
+    p.sync_and_drop("e1")
+    p.clear()
+
+
+def test_non_tensor_values_ignored():
+    p = StreamPlacer()
+    p.record_output("e2", {"not": "a tensor"})
+    p.wait_value("e2", [1, 2, 3])
+    assert "e2" not in p._events
+
+
+def test_workflow_on_cpu_unchanged(tmp_path, monkeypatch):
+    monkeypatch.setenv("LZY_AMD_STORAGE", str(tmp_path / "s"))
+    from lzy_amd import Lzy, op
+    from lzy_amd.runtime.local import LocalRuntime
+
+    @op
+    def add(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+        return a + b
+
+    with Lzy(runtime=LocalRuntime()).workflow("wf", interactive=False):
+        r = add(torch.ones(8), torch.full((8,), 2.0))
+        assert float(r.sum()) == 24.0
