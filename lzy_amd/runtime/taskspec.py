@@ -67,7 +67,11 @@ class WorkerStore:
             self.pickled[entry_id] = pickled
 
     def get(self, entry_id: str) -> Any:
-        return self.values[entry_id]
+        value = self.values[entry_id]
+        from lzy_amd.runtime.streams import STREAMS
+
+        STREAMS.wait_value(entry_id, value)  # order after producing stream
+        return value
 
     def has(self, entry_id: str) -> bool:
         return entry_id in self.values
@@ -75,6 +79,9 @@ class WorkerStore:
     def drop(self, entry_id: str) -> None:
         self.values.pop(entry_id, None)
         self.pickled.pop(entry_id, None)
+        from lzy_amd.runtime.streams import STREAMS
+
+        STREAMS.drop(entry_id)
 
     def clear(self) -> None:
         self.values.clear()
@@ -158,7 +165,20 @@ def run_taskspec(
                 ranks=tuple(spec.gang["ranks"]),
                 process_group=gang_group,
             ))
-        result = func(*args, **kwargs)
+        # per-task HIP stream: back-to-back tasks on this rank overlap on
+        # device; input hand-off is event-ordered (runtime/streams.py)
+        from contextlib import nullcontext
+
+        from lzy_amd.runtime.streams import STREAMS
+
+        stream = STREAMS.next_stream()
+        with (torch.cuda.stream(stream) if stream is not None else nullcontext()):
+            for eid, v in zip(
+                list(spec.arg_entries) + list(spec.kwarg_entries.values()),
+                args + list(kwargs.values()),
+            ):
+                STREAMS.wait_value(eid, v)
+            result = func(*args, **kwargs)
     except BaseException as e:  # noqa: BLE001 - transported as a value
         return _fail(spec, t0, e, traceback.format_exc())
     finally:
@@ -196,11 +216,15 @@ def run_taskspec(
         if not isinstance(value, torch.Tensor):
             pickled = pickle_value(value)
         store.put(eid, value, pickled=pickled)
+        from lzy_amd.runtime.streams import STREAMS as _S
+
+        _S.record_output(eid, value, stream=stream)
         meta = describe_value(eid, value)
         if pickled is not None:
             meta.nbytes = len(pickled)
         outputs.append(meta.to_wire())
         if spec.cache and gang_primary:
+            _S.wait_value(eid, value)  # D2H serialize orders after the op
             data, fmt = serializers.dumps(value)
             storage.write_bytes(out_uris[eid], data)
             _write_fmt(storage, out_uris[eid], fmt)
