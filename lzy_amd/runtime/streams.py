@@ -42,6 +42,7 @@ class StreamPlacer:
         self._streams: Dict[int, list] = {}  # device index -> [Stream]
         self._rr: Dict[int, int] = {}
         self._events: Dict[str, Tuple[Any, Any]] = {}  # entry -> (event, stream)
+        self._tls = threading.local()
         self.enabled = torch.cuda.is_available()
 
     def _pool(self, device: int) -> list:
@@ -57,15 +58,28 @@ class StreamPlacer:
             return self._streams[device]
 
     def next_stream(self) -> Optional[torch.cuda.Stream]:
-        """Stream for the next op on the current device (None on CPU)."""
+        """Stream for ops on the *current thread* (None on CPU).
+
+        Streams are bound per executor thread, round-robin at first use —
+        NOT rotated per op: sequential ops in one thread share a stream
+        (torch's caching allocator is per-stream; rotation fragments it
+        and un-orders hidden op state like persistent model/optimizer
+        tensors), while genuinely concurrent executor threads land on
+        distinct streams and overlap."""
         if not self.enabled:
             return None
         dev = torch.cuda.current_device()
-        pool = self._pool(dev)
-        with self._lock:
-            i = self._rr[dev]
-            self._rr[dev] = (i + 1) % len(pool)
-        return pool[i]
+        by_dev = getattr(self._tls, "stream_by_dev", None)
+        if by_dev is None:
+            by_dev = self._tls.stream_by_dev = {}
+        s = by_dev.get(dev)
+        if s is None:
+            pool = self._pool(dev)
+            with self._lock:
+                i = self._rr[dev]
+                self._rr[dev] = (i + 1) % len(pool)
+            s = by_dev[dev] = pool[i]
+        return s
 
     # -- producer side ------------------------------------------------------
 
