@@ -183,9 +183,17 @@ class DefaultSnapshot:
         STREAMS.wait_value(entry_id, value)  # D2H copy orders after producer
         data, fmt = self._serializers.dumps(value)
         entry.data_format = fmt
+        h = _hash_bytes(data)
+        # content-hash dedup (reference: snapshot md5 dedup before S3
+        # upload, pylzy api/v1/snapshot.py:141-160): identical bytes
+        # already at the URI -> skip the write
+        if self._hashes.get(entry_id) == h and self._storage.blob_exists(
+            entry.storage_uri
+        ):
+            return entry.storage_uri
         self._storage.write_bytes(entry.storage_uri, data)
         self._storage.write_bytes(entry.storage_uri + ".fmt", fmt.encode())
-        self._hashes[entry_id] = _hash_bytes(data)
+        self._hashes[entry_id] = h
         return entry.storage_uri
 
     def load(self, entry_id: str) -> Any:

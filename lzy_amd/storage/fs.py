@@ -31,19 +31,36 @@ class FsStorageClient(StorageClient):
             shutil.copyfileobj(f, dest, _CHUNK)
 
     def write(self, uri: str, data: BinaryIO) -> None:
-        path = uri_to_path(uri)
-        path.parent.mkdir(parents=True, exist_ok=True)
-        fd, tmp = tempfile.mkstemp(dir=path.parent, prefix=".lzy_tmp_")
-        try:
-            with os.fdopen(fd, "wb") as f:
-                shutil.copyfileobj(data, f, _CHUNK)
-            os.replace(tmp, path)
-        except BaseException:
+        """Atomic (tmp + rename), retried on transient OSErrors when the
+        source is rewindable (reference: util-db withRetries around every
+        persist — DbHelper.java:25-33)."""
+        from lzy_amd.utils.retry import with_retries
+
+        start = data.tell() if data.seekable() else None
+
+        def go() -> None:
+            if start is not None:
+                data.seek(start)
+            path = uri_to_path(uri)
+            path.parent.mkdir(parents=True, exist_ok=True)
+            fd, tmp = tempfile.mkstemp(dir=path.parent, prefix=".lzy_tmp_")
             try:
-                os.unlink(tmp)
-            except OSError:
-                pass
-            raise
+                with os.fdopen(fd, "wb") as f:
+                    shutil.copyfileobj(data, f, _CHUNK)
+                os.replace(tmp, path)
+            except BaseException:
+                try:
+                    os.unlink(tmp)
+                except OSError:
+                    pass
+                raise
+
+        with_retries(
+            go,
+            attempts=5 if start is not None else 1,
+            retry_on=(OSError,),
+            what=f"write {uri}",
+        )
 
     def blob_exists(self, uri: str) -> bool:
         return uri_to_path(uri).is_file()
