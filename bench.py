@@ -76,6 +76,12 @@ def ingest(shard_idx: int, seed: int) -> torch.Tensor:
 
 @op
 def preprocess(t: torch.Tensor) -> torch.Tensor:
+    if t.is_cuda:
+        # fused two-pass normalize (HIP): 3.2 GB of HBM traffic per GiB
+        # shard vs ~14 GB for the unfused torch chain
+        from lzy_amd.ops import normalize
+
+        return normalize(t)
     x = t.float()
     x = (x - x.mean()) / (x.std() + 1e-6)
     return x.to(torch.bfloat16)
@@ -83,6 +89,10 @@ def preprocess(t: torch.Tensor) -> torch.Tensor:
 
 @op
 def augment(t: torch.Tensor) -> torch.Tensor:
+    if t.is_cuda:
+        from lzy_amd.ops import scale_shift
+
+        return scale_shift(t, 1.0009765625, 0.125)
     return t * 1.0009765625 + 0.125
 
 
@@ -135,11 +145,19 @@ def checksum(t: torch.Tensor) -> int:
 
 @op
 def merge(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    if a.is_cuda and b.is_cuda:
+        from lzy_amd.ops import axpby
+
+        return axpby(a, b, 0.5, 0.5)
     return (a.float() + b.float()).mul_(0.5).to(torch.bfloat16)
 
 
 @op
 def evaluate(t: torch.Tensor) -> float:
+    if t.is_cuda:
+        from lzy_amd.ops import abs_mean
+
+        return abs_mean(t)
     return float(t.float().abs().mean().item())
 
 

@@ -64,6 +64,27 @@ def _try_load() -> Optional[ctypes.CDLL]:
     lib.lz_fill_pattern.argtypes = [
         ctypes.c_void_p, ctypes.c_int64, ctypes.c_uint64, ctypes.c_void_p,
     ]
+    lib.lz_stats.restype = ctypes.c_int
+    lib.lz_stats.argtypes = [
+        ctypes.c_void_p, ctypes.c_int, ctypes.c_int64, ctypes.c_int,
+        ctypes.c_void_p, ctypes.c_void_p,
+    ]
+    lib.lz_normalize_apply.restype = ctypes.c_int
+    lib.lz_normalize_apply.argtypes = [
+        ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p, ctypes.c_int,
+        ctypes.c_int64, ctypes.c_void_p, ctypes.c_float, ctypes.c_void_p,
+    ]
+    lib.lz_scale_shift.restype = ctypes.c_int
+    lib.lz_scale_shift.argtypes = [
+        ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p, ctypes.c_int,
+        ctypes.c_int64, ctypes.c_float, ctypes.c_float, ctypes.c_void_p,
+    ]
+    lib.lz_axpby.restype = ctypes.c_int
+    lib.lz_axpby.argtypes = [
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p,
+        ctypes.c_int, ctypes.c_int64, ctypes.c_float, ctypes.c_float,
+        ctypes.c_void_p,
+    ]
     lib.lz_error_name.restype = ctypes.c_char_p
     lib.lz_error_name.argtypes = [ctypes.c_int]
     lib.lz_set_max_blocks.restype = None
@@ -176,6 +197,134 @@ def device_checksum(t, method: str = "auto") -> int:
         )
     )
     return int(out.item()) & 0xFFFFFFFFFFFFFFFF
+
+
+def stats(t, use_abs: bool = False):
+    """One-pass (sum, sumsq) — or (sum|x|, sumsq) — of a device tensor.
+    Returns a device float64[2] tensor; no host sync."""
+    import torch
+
+    lib = _require_native()
+    if not t.is_cuda:
+        raise ValueError("stats requires a device tensor")
+    flat = t.detach().contiguous()
+    out = torch.empty(2, dtype=torch.float64, device=t.device)
+    _check(
+        lib.lz_stats(
+            ctypes.c_void_p(flat.data_ptr()),
+            _dtype_code(flat.dtype),
+            flat.numel(),
+            1 if use_abs else 0,
+            ctypes.c_void_p(out.data_ptr()),
+            ctypes.c_void_p(_current_stream_ptr()),
+        )
+    )
+    return out
+
+
+def abs_mean(t) -> float:
+    """Fused mean(|x|) — one read of the buffer, 8 bytes back to host.
+    (torch chain `t.float().abs().mean()` moves ~9x the bytes.)"""
+    s = stats(t, use_abs=True)
+    return float(s[0].item()) / max(1, t.numel())
+
+
+def mean_std(t, unbiased: bool = False):
+    """(mean, std) from the one-pass stats kernel; two scalars back."""
+    import math
+
+    s = stats(t, use_abs=False)
+    n = t.numel()
+    vals = s.cpu()
+    mean = float(vals[0]) / n
+    var = float(vals[1]) / n - mean * mean
+    if unbiased and n > 1:
+        var *= n / (n - 1)
+    return mean, math.sqrt(max(var, 0.0))
+
+
+def normalize(src, dst=None, eps: float = 1e-6):
+    """dst = (src - mean(src)) / (std(src) + ~eps), fused two-pass on
+    device (stats kernel + apply kernel, stats handed over in HBM — no
+    host round-trip).  dst defaults to a new tensor of src's dtype."""
+    import torch
+
+    lib = _require_native()
+    if not src.is_cuda:
+        raise ValueError("normalize requires a device tensor")
+    s = src.detach().contiguous()
+    if dst is None:
+        dst = torch.empty_like(s)
+    st = stats(s, use_abs=False)
+    _check(
+        lib.lz_normalize_apply(
+            ctypes.c_void_p(s.data_ptr()),
+            _dtype_code(s.dtype),
+            ctypes.c_void_p(dst.data_ptr()),
+            _dtype_code(dst.dtype),
+            s.numel(),
+            ctypes.c_void_p(st.data_ptr()),
+            ctypes.c_float(eps),
+            ctypes.c_void_p(_current_stream_ptr()),
+        )
+    )
+    return dst
+
+
+def scale_shift(src, a: float, b: float, dst=None):
+    """dst = src * a + b (fused single-FMA elementwise; in-place when
+    dst is src)."""
+    import torch
+
+    lib = _require_native()
+    if not src.is_cuda:
+        raise ValueError("scale_shift requires a device tensor")
+    s = src.detach().contiguous()
+    if dst is None:
+        dst = torch.empty_like(s)
+    _check(
+        lib.lz_scale_shift(
+            ctypes.c_void_p(s.data_ptr()),
+            _dtype_code(s.dtype),
+            ctypes.c_void_p(dst.data_ptr()),
+            _dtype_code(dst.dtype),
+            s.numel(),
+            ctypes.c_float(a),
+            ctypes.c_float(b),
+            ctypes.c_void_p(_current_stream_ptr()),
+        )
+    )
+    return dst
+
+
+def axpby(a, b, alpha: float = 1.0, beta: float = 1.0, dst=None):
+    """dst = alpha*a + beta*b, fused (the tree-merge primitive: one read
+    of each input, one write, vs 4+ passes for the torch float() chain)."""
+    import torch
+
+    lib = _require_native()
+    if not (a.is_cuda and b.is_cuda):
+        raise ValueError("axpby requires device tensors")
+    if a.numel() != b.numel() or a.dtype != b.dtype:
+        raise ValueError("axpby: shape/dtype mismatch")
+    x = a.detach().contiguous()
+    y = b.detach().contiguous()
+    if dst is None:
+        dst = torch.empty_like(x)
+    _check(
+        lib.lz_axpby(
+            ctypes.c_void_p(x.data_ptr()),
+            ctypes.c_void_p(y.data_ptr()),
+            _dtype_code(x.dtype),
+            ctypes.c_void_p(dst.data_ptr()),
+            _dtype_code(dst.dtype),
+            x.numel(),
+            ctypes.c_float(alpha),
+            ctypes.c_float(beta),
+            ctypes.c_void_p(_current_stream_ptr()),
+        )
+    )
+    return dst
 
 
 def fill_pattern(t, seed: int = 0) -> None:

@@ -336,6 +336,276 @@ extern "C" hipError_t lz_checksum_mfma(const void* data, int64_t nbytes,
 }
 
 // ---------------------------------------------------------------------------
+// Fused elementwise/reduction ops — the "standard op" toolkit for workflow
+// stages.  Rationale (MI355X performance rule: HBM traffic is the budget):
+// an unfused torch chain like `(x.float() - x.mean()) / x.std()` moves
+// ~14 GB per 1 GiB bf16 shard across 6 kernels; the fused pair below moves
+// 3.2 GB in 2.  All kernels: vectorized 8 elems/lane, persistent grids,
+// f32 lane accumulation folded to f64 per wave (one f64 atomic per wave).
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ double lz_wave_reduce_f64(double v) {
+    for (int off = 32; off > 0; off >>= 1)
+        v += __shfl_down(v, off, 64);
+    return v;
+}
+
+// sum of x and x^2 (or |x| with ABS=1) in one pass; out[0]+=sum, out[1]+=sumsq
+template <typename T, bool ABS>
+__global__ void stats_kernel(const T* __restrict__ src, int64_t n,
+                             double* __restrict__ out) {
+    constexpr int V = 8;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    int64_t vec_n = n / V;
+    using SrcV = struct { T v[V]; };
+    const SrcV* srcv = reinterpret_cast<const SrcV*>(src);
+
+    float s = 0.f, s2 = 0.f;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < vec_n;
+         i += stride) {
+        SrcV x = srcv[i];
+#pragma unroll
+        for (int k = 0; k < V; ++k) {
+            float f = lz_to_float<T>(x.v[k]);
+            s += ABS ? fabsf(f) : f;
+            s2 += f * f;
+        }
+    }
+    int64_t tail_start = vec_n * V;
+    for (int64_t i = tail_start + (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < n; i += stride) {
+        float f = lz_to_float<T>(src[i]);
+        s += ABS ? fabsf(f) : f;
+        s2 += f * f;
+    }
+    double ds = lz_wave_reduce_f64((double)s);
+    double ds2 = lz_wave_reduce_f64((double)s2);
+    if ((threadIdx.x & 63) == 0) {
+        atomicAdd(&out[0], ds);
+        atomicAdd(&out[1], ds2);
+    }
+}
+
+// dst = cast((src - mean) * rstd): stats read from the device buffer the
+// stats kernel filled — no host round-trip between the two passes.
+template <typename SrcT, typename DstT>
+__global__ void normalize_apply_kernel(const SrcT* __restrict__ src,
+                                       DstT* __restrict__ dst, int64_t n,
+                                       const double* __restrict__ stats,
+                                       float eps) {
+    const float mean = (float)(stats[0] / (double)n);
+    const float var = (float)(stats[1] / (double)n) - mean * mean;
+    const float rstd = rsqrtf(fmaxf(var, 0.f) + eps);
+
+    constexpr int V = 8;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    int64_t vec_n = n / V;
+    using SrcV = struct { SrcT v[V]; };
+    using DstV = struct { DstT v[V]; };
+    const SrcV* srcv = reinterpret_cast<const SrcV*>(src);
+    DstV* dstv = reinterpret_cast<DstV*>(dst);
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < vec_n;
+         i += stride) {
+        SrcV x = srcv[i];
+        DstV d;
+#pragma unroll
+        for (int k = 0; k < V; ++k)
+            d.v[k] = lz_from_float<DstT>((lz_to_float<SrcT>(x.v[k]) - mean) * rstd);
+        dstv[i] = d;
+    }
+    int64_t tail_start = vec_n * V;
+    for (int64_t i = tail_start + (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < n; i += stride)
+        dst[i] = lz_from_float<DstT>((lz_to_float<SrcT>(src[i]) - mean) * rstd);
+}
+
+// dst = cast(src * a + b) — fused affine (one FMA per element)
+template <typename SrcT, typename DstT>
+__global__ void scale_shift_kernel(const SrcT* __restrict__ src,
+                                   DstT* __restrict__ dst, int64_t n,
+                                   float a, float b) {
+    constexpr int V = 8;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    int64_t vec_n = n / V;
+    using SrcV = struct { SrcT v[V]; };
+    using DstV = struct { DstT v[V]; };
+    const SrcV* srcv = reinterpret_cast<const SrcV*>(src);
+    DstV* dstv = reinterpret_cast<DstV*>(dst);
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < vec_n;
+         i += stride) {
+        SrcV x = srcv[i];
+        DstV d;
+#pragma unroll
+        for (int k = 0; k < V; ++k)
+            d.v[k] = lz_from_float<DstT>(fmaf(lz_to_float<SrcT>(x.v[k]), a, b));
+        dstv[i] = d;
+    }
+    int64_t tail_start = vec_n * V;
+    for (int64_t i = tail_start + (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < n; i += stride)
+        dst[i] = lz_from_float<DstT>(fmaf(lz_to_float<SrcT>(src[i]), a, b));
+}
+
+// dst = cast(alpha*a + beta*b) — fused binary combine (tree-merge stages)
+template <typename SrcT, typename DstT>
+__global__ void axpby_kernel(const SrcT* __restrict__ a,
+                             const SrcT* __restrict__ b,
+                             DstT* __restrict__ dst, int64_t n, float alpha,
+                             float beta) {
+    constexpr int V = 8;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    int64_t vec_n = n / V;
+    using SrcV = struct { SrcT v[V]; };
+    using DstV = struct { DstT v[V]; };
+    const SrcV* av = reinterpret_cast<const SrcV*>(a);
+    const SrcV* bv = reinterpret_cast<const SrcV*>(b);
+    DstV* dstv = reinterpret_cast<DstV*>(dst);
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < vec_n;
+         i += stride) {
+        SrcV x = av[i], y = bv[i];
+        DstV d;
+#pragma unroll
+        for (int k = 0; k < V; ++k)
+            d.v[k] = lz_from_float<DstT>(
+                fmaf(lz_to_float<SrcT>(x.v[k]), alpha,
+                     lz_to_float<SrcT>(y.v[k]) * beta));
+        dstv[i] = d;
+    }
+    int64_t tail_start = vec_n * V;
+    for (int64_t i = tail_start + (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < n; i += stride)
+        dst[i] = lz_from_float<DstT>(
+            fmaf(lz_to_float<SrcT>(a[i]), alpha, lz_to_float<SrcT>(b[i]) * beta));
+}
+
+static inline int lz_grid_for(int64_t work_items) {
+    int64_t want = (work_items + LZ_BLOCK - 1) / LZ_BLOCK;
+    return (int)(want < 1 ? 1 : (want > g_max_blocks ? g_max_blocks : want));
+}
+
+// out_device: double[2], zeroed here; out[0]=sum(|x| if abs else x), out[1]=sum(x^2)
+extern "C" hipError_t lz_stats(const void* src, int dtype, int64_t n, int use_abs,
+                               double* out_device, void* stream) {
+    hipStream_t s = (hipStream_t)stream;
+    hipError_t err = hipMemsetAsync(out_device, 0, 16, s);
+    if (err != hipSuccess) return err;
+    int blocks = lz_grid_for(n / 8);
+#define LZ_STATS_CASE(code, T)                                                  \
+    case code:                                                                  \
+        if (use_abs)                                                            \
+            hipLaunchKernelGGL((stats_kernel<T, true>), dim3(blocks),           \
+                               dim3(LZ_BLOCK), 0, s, (const T*)src, n,          \
+                               out_device);                                     \
+        else                                                                    \
+            hipLaunchKernelGGL((stats_kernel<T, false>), dim3(blocks),          \
+                               dim3(LZ_BLOCK), 0, s, (const T*)src, n,          \
+                               out_device);                                     \
+        break;
+    switch (dtype) {
+        LZ_STATS_CASE(LZ_F32, float)
+        LZ_STATS_CASE(LZ_F16, __half)
+        LZ_STATS_CASE(LZ_BF16, __hip_bfloat16)
+        default: return hipErrorInvalidValue;
+    }
+#undef LZ_STATS_CASE
+    return hipGetLastError();
+}
+
+extern "C" hipError_t lz_normalize_apply(const void* src, int src_dtype,
+                                         void* dst, int dst_dtype, int64_t n,
+                                         const double* stats_device, float eps,
+                                         void* stream) {
+    hipStream_t s = (hipStream_t)stream;
+    int blocks = lz_grid_for(n / 8);
+#define LZ_NA_DST(SrcT, code, DstT)                                             \
+    case code:                                                                  \
+        hipLaunchKernelGGL((normalize_apply_kernel<SrcT, DstT>), dim3(blocks),  \
+                           dim3(LZ_BLOCK), 0, s, (const SrcT*)src, (DstT*)dst,  \
+                           n, stats_device, eps);                               \
+        break;
+#define LZ_NA_SRC(scode, SrcT)                                                  \
+    case scode:                                                                 \
+        switch (dst_dtype) {                                                    \
+            LZ_NA_DST(SrcT, LZ_F32, float)                                      \
+            LZ_NA_DST(SrcT, LZ_F16, __half)                                     \
+            LZ_NA_DST(SrcT, LZ_BF16, __hip_bfloat16)                            \
+            default: return hipErrorInvalidValue;                               \
+        }                                                                       \
+        break;
+    switch (src_dtype) {
+        LZ_NA_SRC(LZ_F32, float)
+        LZ_NA_SRC(LZ_F16, __half)
+        LZ_NA_SRC(LZ_BF16, __hip_bfloat16)
+        default: return hipErrorInvalidValue;
+    }
+#undef LZ_NA_SRC
+#undef LZ_NA_DST
+    return hipGetLastError();
+}
+
+extern "C" hipError_t lz_scale_shift(const void* src, int src_dtype, void* dst,
+                                     int dst_dtype, int64_t n, float a, float b,
+                                     void* stream) {
+    hipStream_t s = (hipStream_t)stream;
+    int blocks = lz_grid_for(n / 8);
+#define LZ_SS_DST(SrcT, code, DstT)                                             \
+    case code:                                                                  \
+        hipLaunchKernelGGL((scale_shift_kernel<SrcT, DstT>), dim3(blocks),      \
+                           dim3(LZ_BLOCK), 0, s, (const SrcT*)src, (DstT*)dst,  \
+                           n, a, b);                                            \
+        break;
+#define LZ_SS_SRC(scode, SrcT)                                                  \
+    case scode:                                                                 \
+        switch (dst_dtype) {                                                    \
+            LZ_SS_DST(SrcT, LZ_F32, float)                                      \
+            LZ_SS_DST(SrcT, LZ_F16, __half)                                     \
+            LZ_SS_DST(SrcT, LZ_BF16, __hip_bfloat16)                            \
+            default: return hipErrorInvalidValue;                               \
+        }                                                                       \
+        break;
+    switch (src_dtype) {
+        LZ_SS_SRC(LZ_F32, float)
+        LZ_SS_SRC(LZ_F16, __half)
+        LZ_SS_SRC(LZ_BF16, __hip_bfloat16)
+        default: return hipErrorInvalidValue;
+    }
+#undef LZ_SS_SRC
+#undef LZ_SS_DST
+    return hipGetLastError();
+}
+
+extern "C" hipError_t lz_axpby(const void* a, const void* b, int src_dtype,
+                               void* dst, int dst_dtype, int64_t n, float alpha,
+                               float beta, void* stream) {
+    hipStream_t s = (hipStream_t)stream;
+    int blocks = lz_grid_for(n / 8);
+#define LZ_AX_DST(SrcT, code, DstT)                                             \
+    case code:                                                                  \
+        hipLaunchKernelGGL((axpby_kernel<SrcT, DstT>), dim3(blocks),            \
+                           dim3(LZ_BLOCK), 0, s, (const SrcT*)a,                \
+                           (const SrcT*)b, (DstT*)dst, n, alpha, beta);         \
+        break;
+#define LZ_AX_SRC(scode, SrcT)                                                  \
+    case scode:                                                                 \
+        switch (dst_dtype) {                                                    \
+            LZ_AX_DST(SrcT, LZ_F32, float)                                      \
+            LZ_AX_DST(SrcT, LZ_F16, __half)                                     \
+            LZ_AX_DST(SrcT, LZ_BF16, __hip_bfloat16)                            \
+            default: return hipErrorInvalidValue;                               \
+        }                                                                       \
+        break;
+    switch (src_dtype) {
+        LZ_AX_SRC(LZ_F32, float)
+        LZ_AX_SRC(LZ_F16, __half)
+        LZ_AX_SRC(LZ_BF16, __hip_bfloat16)
+        default: return hipErrorInvalidValue;
+    }
+#undef LZ_AX_SRC
+#undef LZ_AX_DST
+    return hipGetLastError();
+}
+
+// ---------------------------------------------------------------------------
 // fill_pattern: test/verification helper (deterministic device-side fill).
 // ---------------------------------------------------------------------------
 

@@ -197,3 +197,91 @@ def test_checksum_mfma_bandwidth():
     gbps = t.numel() * 4 / dt / 1e9
     print(f"device_checksum mfma: {gbps:.0f} GB/s")
     assert gbps > 1500, f"mfma checksum too slow: {gbps:.0f} GB/s"
+
+
+@requires_gpu
+def test_stats_and_abs_mean_numerics():
+    """stats/abs_mean vs the plain torch fp32 reference."""
+    from lzy_amd.ops import abs_mean, mean_std, stats
+
+    for n in [1, 7, 1024, (1 << 20) + 3]:
+        t = torch.randn(n, device="cuda", dtype=torch.float32)
+        ref = t.double()
+        s = stats(t).cpu()
+        assert abs(float(s[0]) - float(ref.sum())) <= 1e-3 * max(1.0, abs(float(ref.sum())))
+        assert abs(float(s[1]) - float((ref * ref).sum())) <= 1e-3 * float((ref * ref).sum() + 1)
+        am = abs_mean(t)
+        assert abs(am - float(ref.abs().mean())) < 1e-4
+        m, sd = mean_std(t)
+        assert abs(m - float(ref.mean())) < 1e-4
+        assert abs(sd - float(ref.std(unbiased=False))) < 1e-3
+
+    tb = torch.randn(1 << 20, device="cuda").to(torch.bfloat16)
+    am = abs_mean(tb)
+    ref = float(tb.float().abs().mean())
+    assert abs(am - ref) < 1e-3
+
+
+@requires_gpu
+def test_normalize_numerics():
+    from lzy_amd.ops import normalize
+
+    t = (torch.randn(1 << 20, device="cuda") * 3 + 5).to(torch.bfloat16)
+    out = normalize(t)
+    assert out.dtype == torch.bfloat16
+    ref_x = t.float()
+    ref = (ref_x - ref_x.mean()) / (ref_x.std(unbiased=False) + 0)
+    diff = (out.float() - ref).abs().max().item()
+    assert diff < 0.05, f"normalize mismatch {diff}"
+    # normalized output: mean ~0, std ~1
+    assert abs(float(out.float().mean())) < 1e-2
+    assert abs(float(out.float().std()) - 1.0) < 1e-2
+
+
+@requires_gpu
+def test_scale_shift_and_axpby_numerics():
+    from lzy_amd.ops import axpby, scale_shift
+
+    t = torch.randn(12345, device="cuda", dtype=torch.float32)
+    out = scale_shift(t, 2.5, -1.25)
+    ref = t * 2.5 - 1.25
+    assert torch.allclose(out, ref, atol=1e-6)
+
+    a = torch.randn(99991, device="cuda").to(torch.bfloat16)
+    b = torch.randn(99991, device="cuda").to(torch.bfloat16)
+    out = axpby(a, b, 0.5, 0.5)
+    ref = ((a.float() + b.float()) * 0.5).to(torch.bfloat16)
+    assert (out.float() - ref.float()).abs().max().item() < 0.02
+
+
+@requires_gpu
+def test_fused_kernels_bandwidth():
+    import time
+
+    from lzy_amd.ops import abs_mean, normalize
+
+    t = torch.empty(256 << 20, device="cuda", dtype=torch.bfloat16)  # 512 MiB
+    t.normal_()
+    nbytes = t.numel() * 2
+
+    abs_mean(t)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(5):
+        abs_mean(t)
+    dt = (time.perf_counter() - t0) / 5
+    gbps = nbytes / dt / 1e9
+    print(f"abs_mean: {gbps:.0f} GB/s")
+    assert gbps > 1500, f"abs_mean too slow: {gbps:.0f} GB/s"
+
+    out = torch.empty_like(t)
+    normalize(t, dst=out)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(5):
+        normalize(t, dst=out)
+    torch.cuda.synchronize()
+    dt = (time.perf_counter() - t0) / 5
+    gbps = 3 * nbytes / dt / 1e9  # 2 reads + 1 write
+    print(f"normalize: {gbps:.0f} GB/s effective")
+    assert gbps > 2000, f"normalize too slow: {gbps:.0f} GB/s"
