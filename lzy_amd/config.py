@@ -54,6 +54,9 @@ class Config:
     heartbeat_period_s: float = 2.0    # worker liveness probe period
     # result cache / snapshot
     cache_enabled: bool = True
+    # HBM store spill tier (pinned-host async)
+    spill_enabled: bool = True
+    spill_threshold_frac: float = 0.85
     # logs / metrics
     log_archive: bool = True           # archive per-op std logs to durable tier
     metrics_port: int = 0              # >0 -> serve /metrics on this port

@@ -97,7 +97,7 @@ class WorkerAgent:
         self.rank = rank
         self.world = world
         self.device = device
-        self.store = WorkerStore()
+        self.store = WorkerStore(device=device)
         self.serializers = LzySerializerRegistry()
         self.storage = FsStorageClient()
         self.transport = Transport(pg_data, device, world=world)

@@ -55,19 +55,32 @@ class TaskResult:
 class WorkerStore:
     """Per-rank value store: entry_id -> live python object (device tensors
     stay in HBM).  Non-tensor values keep a pre-pickled byte image so
-    transfer sizes are known up front (single-phase recv)."""
+    transfer sizes are known up front (single-phase recv).
 
-    def __init__(self) -> None:
+    Past the HBM threshold, LRU device tensors spill to pinned host
+    memory and come back transparently at get() — the 288 GB-per-GPU
+    result store with pinned-host async spill (storage/spill.py)."""
+
+    def __init__(self, device=None) -> None:
         self.values: Dict[str, Any] = {}
         self.pickled: Dict[str, bytes] = {}
+        from lzy_amd.storage.spill import SpillManager
+
+        self.spill = SpillManager(device=device)
 
     def put(self, entry_id: str, value: Any, pickled: Optional[bytes] = None) -> None:
         self.values[entry_id] = value
         if pickled is not None:
             self.pickled[entry_id] = pickled
+        self.spill.track(entry_id, value)
+        self.spill.maybe_spill(self.values)
 
     def get(self, entry_id: str) -> Any:
+        if self.spill.is_spilled(entry_id):
+            value = self.spill.unspill(entry_id, self.values)
+            return value
         value = self.values[entry_id]
+        self.spill.track(entry_id, value)  # LRU touch
         from lzy_amd.runtime.streams import STREAMS
 
         STREAMS.wait_value(entry_id, value)  # order after producing stream
@@ -79,6 +92,7 @@ class WorkerStore:
     def drop(self, entry_id: str) -> None:
         self.values.pop(entry_id, None)
         self.pickled.pop(entry_id, None)
+        self.spill.forget(entry_id)
         from lzy_amd.runtime.streams import STREAMS
 
         STREAMS.drop(entry_id)

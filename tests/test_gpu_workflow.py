@@ -135,3 +135,38 @@ def test_stream_race_checker_detects_unordered_put(monkeypatch):
     t = torch.ones(16, device="cuda")
     with pytest.raises(RuntimeError, match="stream-race"):
         STREAMS.wait_value("never-recorded-entry", t)
+
+
+@requires_gpu
+def test_pinned_host_spill_roundtrip():
+    """HBM store spill tier: past the (test-lowered) threshold, LRU
+    tensors move to pinned host and come back bit-identical on access."""
+    from lzy_amd.runtime.taskspec import WorkerStore
+    from lzy_amd.storage.spill import SpillManager, _DefaultMover
+
+    dev = torch.device("cuda", 0)
+    store = WorkerStore(device=dev)
+    # force pressure: everything above 0 bytes spills
+    store.spill = SpillManager(
+        device=dev,
+        threshold_frac=0.0,
+        bytes_in_use=lambda: 1,
+        capacity=lambda: 1,
+        mover=_DefaultMover(dev),
+    )
+    a = torch.randn(1 << 20, device=dev)
+    b = torch.randn(1 << 20, device=dev)
+    ref_a, ref_b = a.cpu().clone(), b.cpu().clone()
+    store.put("a", a)
+    assert store.spill.is_spilled("a")
+    host_a = store.values["a"]
+    assert not host_a.is_cuda and host_a.is_pinned()
+    store.put("b", b)
+    assert store.spill.is_spilled("b")
+
+    back_a = store.get("a")
+    assert back_a.is_cuda
+    assert torch.equal(back_a.cpu(), ref_a)
+    back_b = store.get("b")
+    assert torch.equal(back_b.cpu(), ref_b)
+    assert not store.spill.is_spilled("a")
