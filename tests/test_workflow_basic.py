@@ -171,3 +171,26 @@ def test_independent_ops_parallel(lzy):
     with lzy.workflow("wf"):
         rs = [wait_all(i) for i in range(4)]
         assert sorted(int(r) for r in rs) == [0, 1, 2, 3]
+
+
+def test_lazy_arguments_defer_materialization(lzy):
+    """@op(lazy_arguments=True): the op body receives proxies and decides
+    what to touch (reference: startup.py lazy slot reads)."""
+    from lzy_amd.proxy import is_lzy_proxy
+
+    seen = {}
+
+    @op
+    def src(x: int) -> int:
+        return x * 2
+
+    @op(lazy_arguments=True)
+    def probe(a: int, b: int) -> int:
+        seen["a_proxy"] = is_lzy_proxy(a)
+        seen["b_proxy"] = is_lzy_proxy(b)
+        return int(a) + 1  # touch only a
+
+    with lzy.workflow("wf", interactive=False):
+        r = probe(src(5), src(7))
+        assert int(r) == 11
+    assert seen["a_proxy"] and seen["b_proxy"]
