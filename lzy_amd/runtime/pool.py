@@ -772,7 +772,15 @@ class _DriverScheduler:
         if byrank:
             best = max(byrank.items(), key=lambda kv: (kv[1], -self.outstanding[kv[0]]))
             return best[0]
-        return min(self.outstanding, key=lambda r: (self.outstanding[r], r))
+        # no data affinity: least-loaded, ties broken round-robin — a
+        # fixed min-rank tie-break collapses fast-completing fan-outs
+        # onto rank 0 (completions race dispatch; single-task barriers
+        # from eager materialization would otherwise all see a fresh
+        # batch).  The rotation lives on the pool so it survives batches.
+        lo = min(self.outstanding.values())
+        cands = sorted(r for r, o in self.outstanding.items() if o == lo)
+        self.pool.rr_counter = getattr(self.pool, "rr_counter", -1) + 1
+        return cands[self.pool.rr_counter % len(cands)]
 
     def _pick_gang(self, k: int) -> List[int]:
         ranks = sorted(

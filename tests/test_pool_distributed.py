@@ -107,3 +107,9 @@ def test_bench_dag_four_ranks(tmp_path):
     r = _run_distributed("tests/bench_dag_script.py", 4, tmp_path, timeout=240)
     assert r.returncode == 0, r.stdout[-3000:] + r.stderr[-3000:]
     assert "BENCH-DAG-OK" in r.stdout
+
+
+def test_pool_world4_gangs_and_chunks(tmp_path):
+    r = _run_distributed("tests/pool_script_world4.py", 4, tmp_path, timeout=240)
+    assert r.returncode == 0, r.stdout[-3000:] + r.stderr[-3000:]
+    assert "WORLD4-OK" in r.stdout
