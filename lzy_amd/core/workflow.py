@@ -46,6 +46,7 @@ class LzyWorkflow:
         self._call_queue: List["LzyCall"] = []
         self._started = False
         self._finished = False
+        self._confirmed = False
         self._snapshot: Optional[DefaultSnapshot] = None
         self._whiteboards: List[Any] = []  # WritableWhiteboard instances
 
@@ -121,7 +122,31 @@ class LzyWorkflow:
             return
         calls = self._call_queue
         self._call_queue = []
+        self._confirm_execution(calls)
         self.owner.runtime.exec(self, calls)
+
+    def _confirm_execution(self, calls: Sequence["LzyCall"]) -> None:
+        """Interactive confirm before the first graph submission
+        (reference: remote runtime's interactive confirm, api/v1/remote/
+        runtime.py:424).  Only prompts on a TTY; answering no aborts."""
+        if not self.interactive or self._confirmed:
+            return
+        import sys
+
+        if not (hasattr(sys.stdin, "isatty") and sys.stdin.isatty()):
+            self._confirmed = True
+            return
+        names = [c.callable_name for c in calls]
+        preview = ", ".join(names[:8]) + ("..." if len(names) > 8 else "")
+        answer = input(
+            f"Workflow '{self.name}': execute graph of {len(names)} op(s) "
+            f"[{preview}]? (y/n) "
+        ).strip().lower()
+        if answer not in ("y", "yes", ""):
+            raise WorkflowAbortedError(
+                f"workflow {self.name} execution declined by user"
+            )
+        self._confirmed = True
 
     # -- whiteboards ----------------------------------------------------------
 

@@ -194,3 +194,35 @@ def test_lazy_arguments_defer_materialization(lzy):
         r = probe(src(5), src(7))
         assert int(r) == 11
     assert seen["a_proxy"] and seen["b_proxy"]
+
+
+def test_interactive_confirm(lzy, monkeypatch):
+    """interactive=True on a TTY prompts before the first submission;
+    declining aborts the workflow (reference runtime.py:424)."""
+    import sys
+
+    from lzy_amd.exceptions import WorkflowAbortedError
+
+    @op
+    def f(x: int) -> int:
+        return x + 1
+
+    class FakeTty:
+        def isatty(self):
+            return True
+
+    monkeypatch.setattr(sys, "stdin", FakeTty())
+    prompts = []
+
+    monkeypatch.setattr("builtins.input", lambda msg: (prompts.append(msg), "y")[1])
+    with lzy.workflow("ok-wf", interactive=True):
+        assert int(f(1)) == 2
+    assert len(prompts) == 1 and "ok-wf" in prompts[0]
+
+    monkeypatch.setattr("builtins.input", lambda msg: "n")
+    try:
+        with lzy.workflow("no-wf", interactive=True):
+            int(f(1))
+        raise AssertionError("expected WorkflowAbortedError")
+    except WorkflowAbortedError:
+        pass
