@@ -97,6 +97,17 @@ class DefaultSnapshot:
         # optional hook: called on a hot-tier miss to pull the value from a
         # remote owner (GpuPoolRuntime wires this to an xGMI fetch)
         self.fetcher = None
+        # pinned-host spill tier for device tensors (storage/spill.py)
+        from lzy_amd.storage.spill import SpillManager
+
+        import torch as _torch
+
+        dev = (
+            _torch.device("cuda", _torch.cuda.current_device())
+            if _torch.cuda.is_available()
+            else None
+        )
+        self.spill = SpillManager(device=dev)
 
     @property
     def serializers(self) -> LzySerializerRegistry:
@@ -130,10 +141,15 @@ class DefaultSnapshot:
     def put(self, entry_id: str, value: Any) -> None:
         self._values[entry_id] = value
         self._hashes.pop(entry_id, None)
+        self.spill.track(entry_id, value)
+        self.spill.maybe_spill(self._values)
 
     def try_get(self, entry_id: str) -> TryGetResult:
         if entry_id in self._values:
+            if self.spill.is_spilled(entry_id):
+                return TryGetResult(True, self.spill.unspill(entry_id, self._values))
             value = self._values[entry_id]
+            self.spill.track(entry_id, value)  # LRU touch
             # cross-stream hand-off safety: the reader's stream waits on
             # the producing op's completion event (runtime/streams.py)
             from lzy_amd.runtime.streams import STREAMS
@@ -164,6 +180,7 @@ class DefaultSnapshot:
 
     def drop_value(self, entry_id: str) -> None:
         self._values.pop(entry_id, None)
+        self.spill.forget(entry_id)
         from lzy_amd.runtime.streams import STREAMS
 
         STREAMS.drop(entry_id)
