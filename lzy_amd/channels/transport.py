@@ -177,11 +177,17 @@ class Transport:
         starts = range(offset_chunks * per, n, per)
         return [flat[s: min(s + per, n)] for s in starts]
 
-    # -- send ---------------------------------------------------------------
+    # -- op builders ---------------------------------------------------------
+    #
+    # send_ops/recv_ops return (P2POp list, keepalive/finalize) WITHOUT
+    # issuing — callers group ops from several transfers into ONE
+    # dist.batch_isend_irecv (ncclGroupStart/End under RCCL), so
+    # transfers to/from *distinct peers* progress in parallel across
+    # their xGMI links instead of serializing on the comm stream.
 
-    def isend_value(self, value: Any, prepickled: Optional[bytes], dst: int,
-                    offset_chunks: int = 0):
-        """Issue non-blocking send(s); returns (works, keepalive).
+    def send_ops(self, value: Any, prepickled: Optional[bytes], dst: int,
+                 offset_chunks: int = 0):
+        """Build send P2POps; returns (ops, keepalive).
 
         ``offset_chunks`` resumes a partially completed transfer: the
         first ``offset_chunks`` chunks are assumed delivered and skipped.
@@ -194,32 +200,25 @@ class Transport:
                 t = t.cpu()
             per = self._chunk_elems(t.element_size())
             if t.numel() <= per and offset_chunks == 0:
-                work = dist.isend(t, dst=dst, group=self._pg)
-                return [work], t
-            flat = t.view(-1)
-            works = [
-                dist.isend(c, dst=dst, group=self._pg)
-                for c in self._chunks(flat, per, offset_chunks)
-            ]
-            return works, t
+                chunks = [t]
+            else:
+                chunks = self._chunks(t.view(-1), per, offset_chunks)
+            ops = [dist.P2POp(dist.isend, c, dst, group=self._pg) for c in chunks]
+            return ops, t
         data = prepickled if prepickled is not None else pickle_value(value)
         buf = torch.frombuffer(bytearray(data), dtype=torch.uint8)
         per = self._chunk_elems(1)
         if buf.numel() <= per and offset_chunks == 0:
-            work = dist.isend(buf, dst=dst, group=self._pg)
-            return [work], buf
-        works = [
-            dist.isend(c, dst=dst, group=self._pg)
-            for c in self._chunks(buf, per, offset_chunks)
-        ]
-        return works, buf
+            chunks = [buf]
+        else:
+            chunks = self._chunks(buf, per, offset_chunks)
+        ops = [dist.P2POp(dist.isend, c, dst, group=self._pg) for c in chunks]
+        return ops, buf
 
-    # -- recv ---------------------------------------------------------------
-
-    def irecv_value(self, meta: EntryMeta, src: int, offset_chunks: int = 0,
-                    into: Optional[torch.Tensor] = None):
-        """Issue non-blocking recv; returns (works, finalize) where
-        finalize() -> the received value (call after works complete).
+    def recv_ops(self, meta: EntryMeta, src: int, offset_chunks: int = 0,
+                 into: Optional[torch.Tensor] = None):
+        """Build recv P2POps; returns (ops, finalize) where finalize() ->
+        the received value (call after the issued works complete).
 
         ``offset_chunks``/``into`` resume into an existing buffer that
         already holds the first ``offset_chunks`` chunks."""
@@ -232,29 +231,44 @@ class Transport:
             buf = into if into is not None else torch.empty(
                 meta.shape, dtype=dtype, device=dev
             )
-            n = buf.numel()
-            if n <= per and offset_chunks == 0:
-                works = [dist.irecv(buf, src=src, group=self._pg)]
+            if buf.numel() <= per and offset_chunks == 0:
+                chunks = [buf]
             else:
-                flat = buf.view(-1)
-                works = [
-                    dist.irecv(c, src=src, group=self._pg)
-                    for c in self._chunks(flat, per, offset_chunks)
-                ]
+                chunks = self._chunks(buf.view(-1), per, offset_chunks)
+            ops = [dist.P2POp(dist.irecv, c, src, group=self._pg) for c in chunks]
             if want_cuda and not on_device:
                 d = self._device
-                return works, (lambda: buf.to(d, non_blocking=False))
-            return works, (lambda: buf)
+                return ops, (lambda: buf.to(d, non_blocking=False))
+            return ops, (lambda: buf)
         per = self._chunk_elems(1)
         buf = into if into is not None else torch.empty(meta.nbytes, dtype=torch.uint8)
         if buf.numel() <= per and offset_chunks == 0:
-            works = [dist.irecv(buf, src=src, group=self._pg)]
+            chunks = [buf]
         else:
-            works = [
-                dist.irecv(c, src=src, group=self._pg)
-                for c in self._chunks(buf, per, offset_chunks)
-            ]
-        return works, (lambda: unpickle_value(buf.numpy().tobytes()))
+            chunks = self._chunks(buf, per, offset_chunks)
+        ops = [dist.P2POp(dist.irecv, c, src, group=self._pg) for c in chunks]
+        return ops, (lambda: unpickle_value(buf.numpy().tobytes()))
+
+    @staticmethod
+    def issue(ops: list) -> list:
+        """Issue a group of P2POps as one batch; returns the works."""
+        if not ops:
+            return []
+        return dist.batch_isend_irecv(ops)
+
+    # -- single-transfer conveniences ----------------------------------------
+
+    def isend_value(self, value: Any, prepickled: Optional[bytes], dst: int,
+                    offset_chunks: int = 0):
+        """Issue non-blocking send(s); returns (works, keepalive)."""
+        ops, keep = self.send_ops(value, prepickled, dst, offset_chunks)
+        return self.issue(ops), keep
+
+    def irecv_value(self, meta: EntryMeta, src: int, offset_chunks: int = 0,
+                    into: Optional[torch.Tensor] = None):
+        """Issue non-blocking recv; returns (works, finalize)."""
+        ops, fin = self.recv_ops(meta, src, offset_chunks, into)
+        return self.issue(ops), fin
 
     @staticmethod
     def completed_chunks(works: list) -> int:

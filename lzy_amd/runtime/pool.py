@@ -127,18 +127,33 @@ class WorkerAgent:
         cmd = msg["cmd"]
         if cmd == "task":
             self._exec_q.put(msg)
-        elif cmd == "xfer_send":
-            eid = msg["entry"]
-            value = self.store.get(eid)
-            works, keep = self.transport.isend_value(
-                value, self.store.pickled.get(eid), msg["dst"]
-            )
-            self._outbox.append((works, keep))
+        elif cmd == "xfer_send_batch":
+            ops_all: list = []
+            keeps: list = []
+            for item in msg["items"]:
+                eid = item["entry"]
+                value = self.store.get(eid)
+                ops, keep = self.transport.send_ops(
+                    value, self.store.pickled.get(eid), item["dst"]
+                )
+                ops_all += ops
+                keeps.append(keep)
+            works = Transport.issue(ops_all)
+            self._outbox.append((works, keeps))
             self._prune_outbox()
-        elif cmd == "xfer_recv":
-            meta = EntryMeta.from_wire(msg["meta"])
-            works, fin = self.transport.irecv_value(meta, msg["src"])
-            self._pending[msg["entry"]] = (works, fin)
+        elif cmd == "xfer_recv_batch":
+            ops_all = []
+            fins: list = []
+            for item in msg["items"]:
+                meta = EntryMeta.from_wire(item["meta"])
+                ops, fin = self.transport.recv_ops(meta, item["src"])
+                fins.append((item["entry"], len(ops), fin))
+                ops_all += ops
+            works = Transport.issue(ops_all)
+            i = 0
+            for eid, nops, fin in fins:
+                self._pending[eid] = (works[i: i + nops], fin)
+                i += nops
         elif cmd == "ipc_export":
             from lzy_amd.channels.transport import export_ipc
 
@@ -531,10 +546,14 @@ class GpuPoolRuntime(Runtime):
         owner = next(iter(meta.owners - {0}), None)
         if owner is None:
             return
-        pool.driver_ctrl.send(owner, {"cmd": "xfer_send", "entry": entry_id, "dst": 0})
         pool.driver_ctrl.send(
-            0, {"cmd": "xfer_recv", "entry": entry_id, "src": owner,
-                "meta": meta.to_wire()}
+            owner,
+            {"cmd": "xfer_send_batch", "items": [{"entry": entry_id, "dst": 0}]},
+        )
+        pool.driver_ctrl.send(
+            0,
+            {"cmd": "xfer_recv_batch",
+             "items": [{"entry": entry_id, "src": owner, "meta": meta.to_wire()}]},
         )
         tag = f"f{pool.next_seq()}"
         pool.driver_ctrl.send(0, {"cmd": "settle", "entries": [entry_id], "tag": tag})
@@ -675,6 +694,8 @@ class _DriverScheduler:
 
         specs_inline: Dict[str, bytes] = {}
         wait_entries_per_rank: Dict[int, List[str]] = {r: [] for r in ranks}
+        sends_by_owner: Dict[int, List[dict]] = {}
+        recvs_by_rank: Dict[int, List[dict]] = {}
 
         for eid in call.input_entry_ids():
             meta = self.meta.get(eid)
@@ -718,17 +739,25 @@ class _DriverScheduler:
                         )
                         METRICS.inc("lzy_transfers_ipc")
                     else:
-                        pool.driver_ctrl.send(
-                            owner, {"cmd": "xfer_send", "entry": eid, "dst": r}
+                        sends_by_owner.setdefault(owner, []).append(
+                            {"entry": eid, "dst": r}
                         )
-                        pool.driver_ctrl.send(
-                            r, {"cmd": "xfer_recv", "entry": eid, "src": owner,
-                                "meta": meta.to_wire()}
+                        recvs_by_rank.setdefault(r, []).append(
+                            {"entry": eid, "src": owner, "meta": meta.to_wire()}
                         )
                         METRICS.inc("lzy_transfers")
                     wait_entries_per_rank[r].append(eid)
                     meta.owners.add(r)
                     METRICS.inc("lzy_transfer_bytes", meta.nbytes)
+
+        # one grouped send/recv command per rank for this task's transfers:
+        # the worker issues the whole group through ONE batch_isend_irecv
+        # (ncclGroupStart/End), so transfers with distinct peers progress
+        # in parallel over their own xGMI links
+        for owner, items in sends_by_owner.items():
+            pool.driver_ctrl.send(owner, {"cmd": "xfer_send_batch", "items": items})
+        for r, items in recvs_by_rank.items():
+            pool.driver_ctrl.send(r, {"cmd": "xfer_recv_batch", "items": items})
 
         func_bytes = _func_bytes(call.signature.func)
         snap = self.workflow.snapshot
