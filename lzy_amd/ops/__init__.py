@@ -64,6 +64,11 @@ def _try_load() -> Optional[ctypes.CDLL]:
     lib.lz_fill_pattern.argtypes = [
         ctypes.c_void_p, ctypes.c_int64, ctypes.c_uint64, ctypes.c_void_p,
     ]
+    lib.lz_fill_pattern_masked.restype = ctypes.c_int
+    lib.lz_fill_pattern_masked.argtypes = [
+        ctypes.c_void_p, ctypes.c_int64, ctypes.c_uint64, ctypes.c_uint64,
+        ctypes.c_void_p,
+    ]
     lib.lz_stats.restype = ctypes.c_int
     lib.lz_stats.argtypes = [
         ctypes.c_void_p, ctypes.c_int, ctypes.c_int64, ctypes.c_int,
@@ -327,11 +332,26 @@ def axpby(a, b, alpha: float = 1.0, beta: float = 1.0, dst=None):
     return dst
 
 
-def fill_pattern(t, seed: int = 0) -> None:
-    """Deterministic device-side fill (tests / synthetic data)."""
+def fill_pattern(t, seed: int = 0, mask16: int = 0) -> None:
+    """Deterministic device-side fill (tests / synthetic data).
+
+    ``mask16``: AND every 16-bit lane with this mask in the same pass —
+    e.g. 0x3FFF makes a bf16 buffer finite positive synthetic data
+    without a second read+write over HBM."""
     lib = _require_native()
     flat = t.detach().contiguous()
     nbytes = flat.numel() * flat.element_size()
+    if mask16:
+        _check(
+            lib.lz_fill_pattern_masked(
+                ctypes.c_void_p(flat.data_ptr()),
+                nbytes,
+                ctypes.c_uint64(seed),
+                ctypes.c_uint64(mask16),
+                ctypes.c_void_p(_current_stream_ptr()),
+            )
+        )
+        return
     _check(
         lib.lz_fill_pattern(
             ctypes.c_void_p(flat.data_ptr()),

@@ -609,12 +609,18 @@ extern "C" hipError_t lz_axpby(const void* a, const void* b, int src_dtype,
 // fill_pattern: test/verification helper (deterministic device-side fill).
 // ---------------------------------------------------------------------------
 
+// mask!=0: each 16-bit lane is ANDed with it after mixing — e.g. 0x3FFF
+// clamps bf16 exponents so the pattern is finite positive synthetic data
+// (fused here: a separate torch bitwise_and_ pass costs 2x the buffer in
+// HBM traffic — measured 389 us per 1 GiB shard, profiles/
+// kernel_stats_bench_fused.md).
 __global__ void fill_pattern_kernel(uint64_t* __restrict__ data, int64_t nwords,
-                                    uint64_t seed) {
+                                    uint64_t seed, uint64_t mask16) {
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    const uint64_t m = mask16 ? (mask16 * 0x0001000100010001ULL) : ~0ULL;
     for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < nwords;
          i += stride) {
-        data[i] = splitmix64(seed ^ (uint64_t)i);
+        data[i] = splitmix64(seed ^ (uint64_t)i) & m;
     }
 }
 
@@ -624,7 +630,19 @@ extern "C" hipError_t lz_fill_pattern(void* data, int64_t nbytes, uint64_t seed,
     int64_t want = (nwords + LZ_BLOCK - 1) / LZ_BLOCK;
     int blocks = (int)(want < 1 ? 1 : (want > g_max_blocks ? g_max_blocks : want));
     hipLaunchKernelGGL(fill_pattern_kernel, dim3(blocks), dim3(LZ_BLOCK), 0,
-                       (hipStream_t)stream, (uint64_t*)data, nwords, seed);
+                       (hipStream_t)stream, (uint64_t*)data, nwords, seed, 0ULL);
+    return hipGetLastError();
+}
+
+extern "C" hipError_t lz_fill_pattern_masked(void* data, int64_t nbytes,
+                                             uint64_t seed, uint64_t mask16,
+                                             void* stream) {
+    int64_t nwords = nbytes >> 3;
+    int64_t want = (nwords + LZ_BLOCK - 1) / LZ_BLOCK;
+    int blocks = (int)(want < 1 ? 1 : (want > g_max_blocks ? g_max_blocks : want));
+    hipLaunchKernelGGL(fill_pattern_kernel, dim3(blocks), dim3(LZ_BLOCK), 0,
+                       (hipStream_t)stream, (uint64_t*)data, nwords, seed,
+                       mask16 & 0xFFFFULL);
     return hipGetLastError();
 }
 

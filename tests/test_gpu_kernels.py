@@ -285,3 +285,18 @@ def test_fused_kernels_bandwidth():
     gbps = 3 * nbytes / dt / 1e9  # 2 reads + 1 write
     print(f"normalize: {gbps:.0f} GB/s effective")
     assert gbps > 2000, f"normalize too slow: {gbps:.0f} GB/s"
+
+
+@requires_gpu
+def test_fill_pattern_mask_fused():
+    """mask16 fused into the fill pass equals the two-pass reference."""
+    from lzy_amd.ops import fill_pattern
+
+    a = torch.empty(1 << 16, device="cuda", dtype=torch.bfloat16)
+    b = torch.empty(1 << 16, device="cuda", dtype=torch.bfloat16)
+    fill_pattern(a, seed=42, mask16=0x3FFF)
+    fill_pattern(b, seed=42)
+    b.view(torch.int16).bitwise_and_(0x3FFF)
+    assert torch.equal(a.view(torch.int16), b.view(torch.int16))
+    assert torch.isfinite(a.float()).all()
+    assert (a.float() >= 0).all()
