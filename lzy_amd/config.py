@@ -50,6 +50,7 @@ class Config:
     op_streams: int = 4                # HIP streams per device for op overlap
     # scheduler / pool
     dispatch_workers: int = 0          # 0 -> auto (LocalRuntime thread pool size)
+    exec_threads: int = 2              # executor threads per pool rank
     gang_timeout_s: float = 120.0      # gang-allocation wait bound
     heartbeat_period_s: float = 2.0    # worker liveness probe period
     # result cache / snapshot

@@ -107,6 +107,21 @@ class WorkerAgent:
         self._groups: Dict[str, Any] = {}
         self._exec_q: "queue.Queue[dict]" = queue.Queue()
         self._shutdown = False
+        # small task pool per rank: independent tasks of one rank overlap
+        # (each executor thread gets its own HIP stream via
+        # runtime/streams.py).  Gang tasks are serialized DRIVER-side
+        # (one gang inflight at a time) so collective order can never
+        # diverge across ranks.
+        from concurrent.futures import ThreadPoolExecutor
+
+        from lzy_amd.config import get_config
+
+        n_exec = max(1, int(getattr(get_config(), "exec_threads", 2)))
+        self._task_pool = ThreadPoolExecutor(
+            max_workers=n_exec, thread_name_prefix=f"lzy-task-r{rank}"
+        )
+        self._futs_lock = threading.Lock()
+        self._futs: Set[Any] = set()
         self._exec_thread = threading.Thread(
             target=self._executor, daemon=True, name=f"lzy-exec-r{rank}"
         )
@@ -206,16 +221,17 @@ class WorkerAgent:
             msg = self._exec_q.get()
             cmd = msg["cmd"]
             if cmd == "_stop":
+                self._task_pool.shutdown(wait=False)
                 return
-            try:
-                if cmd == "task":
-                    self._run_task(msg)
-                elif cmd == "settle":
-                    self._settle(msg["entries"])
-                    self.ctrl.send_event(
-                        {"ev": "settled", "tag": msg["tag"], "rank": self.rank}
-                    )
-                elif cmd == "barrier":
+            if cmd in ("task", "settle"):
+                fut = self._task_pool.submit(self._run_guarded, msg)
+                with self._futs_lock:
+                    self._futs.add(fut)
+                fut.add_done_callback(self._fut_done)
+            elif cmd == "barrier":
+                # a barrier orders after every previously submitted task
+                self._drain_tasks()
+                try:
                     if self.device is not None:
                         torch.cuda.synchronize(self.device)
                     if dist.is_initialized():
@@ -232,22 +248,57 @@ class WorkerAgent:
                             "ts": time.perf_counter(),
                         }
                     )
-            except BaseException as e:  # noqa: BLE001 - agent must survive
-                _LOG.exception("agent r%d failed handling %s", self.rank, cmd)
+                except BaseException as e:  # noqa: BLE001
+                    self._report_error(msg, e)
+
+    def _fut_done(self, fut) -> None:
+        with self._futs_lock:
+            self._futs.discard(fut)
+
+    def _drain_tasks(self) -> None:
+        while True:
+            with self._futs_lock:
+                futs = list(self._futs)
+            if not futs:
+                return
+            for f in futs:
+                f.exception()  # wait; errors were already reported
+
+    def _run_guarded(self, msg: dict) -> None:
+        cmd = msg["cmd"]
+        try:
+            if cmd == "task":
+                self._run_task(msg)
+            else:
+                self._settle(msg["entries"])
                 self.ctrl.send_event(
-                    {
-                        "ev": "agent_error",
-                        "rank": self.rank,
-                        "error": f"{type(e).__name__}: {e}",
-                        "task_id": msg.get("spec").task_id if msg.get("spec") else None,
-                    }
+                    {"ev": "settled", "tag": msg["tag"], "rank": self.rank}
                 )
+        except BaseException as e:  # noqa: BLE001 - agent must survive
+            _LOG.exception("agent r%d failed handling %s", self.rank, cmd)
+            self._report_error(msg, e)
+
+    def _report_error(self, msg: dict, e: BaseException) -> None:
+        self.ctrl.send_event(
+            {
+                "ev": "agent_error",
+                "rank": self.rank,
+                "error": f"{type(e).__name__}: {e}",
+                "task_id": msg.get("spec").task_id if msg.get("spec") else None,
+            }
+        )
 
     def _settle(self, entries: Sequence[str]) -> None:
         """Complete pending transfers, landing values in the store."""
         for eid in entries:
             pending = self._pending.pop(eid, None)
             if pending is None:
+                # either already settled, or a concurrent task's settle
+                # owns the transfer: wait for the value to land
+                if not self.store.has(eid) and not self.store.wait_present(eid):
+                    raise RuntimeError(
+                        f"rank {self.rank}: entry {eid} never arrived"
+                    )
                 continue
             works, fin = pending
             if works:
@@ -578,6 +629,14 @@ class _DriverScheduler:
         self.gang_pending: Dict[str, Set[int]] = {}
         self.errors: List[BaseException] = []
         self.inflight = 0
+        # entries whose transfer to a rank was initiated in this batch:
+        # with intra-rank task concurrency, every later consumer task on
+        # that rank must also wait for the settle (taskspec.wait_present)
+        self.transferred_now: Set[Tuple[int, str]] = set()
+        # collectives must start in the same order on every rank: at most
+        # one gang task inflight at a time, the rest queue here
+        self.gang_inflight: Optional[str] = None
+        self.deferred_gangs: List[str] = []
 
     # -- metadata helpers ---------------------------------------------------
 
@@ -681,9 +740,13 @@ class _DriverScheduler:
 
         dispatch_t0 = time.perf_counter()
         if gpu_count > 1:
+            if self.gang_inflight is not None:
+                self.deferred_gangs.append(task_id)
+                return
             ranks = self._pick_gang(gpu_count)
             tag = pool.ensure_group(ranks)
             gang = {"ranks": ranks, "tag": tag}
+            self.gang_inflight = task_id
         else:
             ranks = [self._pick_rank(call)]
             gang = None
@@ -747,8 +810,13 @@ class _DriverScheduler:
                         )
                         METRICS.inc("lzy_transfers")
                     wait_entries_per_rank[r].append(eid)
+                    self.transferred_now.add((r, eid))
                     meta.owners.add(r)
                     METRICS.inc("lzy_transfer_bytes", meta.nbytes)
+                elif (r, eid) in self.transferred_now:
+                    # transfer already initiated for an earlier task this
+                    # batch; this task must still wait for its settle
+                    wait_entries_per_rank[r].append(eid)
 
         # one grouped send/recv command per rank for this task's transfers:
         # the worker issues the whole group through ONE batch_isend_irecv
@@ -832,6 +900,7 @@ class _DriverScheduler:
             if gang:
                 return False  # wait for the rest of the gang
             self.gang_pending.pop(result.task_id, None)
+            self._release_gang(result.task_id)
         self.inflight -= 1
         self.journal.record(result.task_id, "done")
         ts = self.task_dispatch_ts.pop(result.task_id, None)
@@ -844,6 +913,12 @@ class _DriverScheduler:
         if result.cached:
             METRICS.inc("lzy_cache_hits_pool")
         return True
+
+    def _release_gang(self, task_id: str) -> None:
+        if self.gang_inflight == task_id:
+            self.gang_inflight = None
+            if self.deferred_gangs:
+                self._dispatch(self.deferred_gangs.pop(0))
 
     def _record_outputs(self, call: "LzyCall", rank: int, result: TaskResult) -> None:
         snap = self.workflow.snapshot
@@ -874,4 +949,5 @@ class _DriverScheduler:
             if gang:
                 return  # remaining gang members still must report
             self.gang_pending.pop(result.task_id, None)
+            self._release_gang(result.task_id)
         self.inflight -= 1

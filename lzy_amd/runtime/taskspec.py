@@ -62,18 +62,31 @@ class WorkerStore:
     result store with pinned-host async spill (storage/spill.py)."""
 
     def __init__(self, device=None) -> None:
+        import threading
+
         self.values: Dict[str, Any] = {}
         self.pickled: Dict[str, bytes] = {}
+        self._cond = threading.Condition()
         from lzy_amd.storage.spill import SpillManager
 
         self.spill = SpillManager(device=device)
 
     def put(self, entry_id: str, value: Any, pickled: Optional[bytes] = None) -> None:
-        self.values[entry_id] = value
-        if pickled is not None:
-            self.pickled[entry_id] = pickled
+        with self._cond:
+            self.values[entry_id] = value
+            if pickled is not None:
+                self.pickled[entry_id] = pickled
+            self._cond.notify_all()
         self.spill.track(entry_id, value)
         self.spill.maybe_spill(self.values)
+
+    def wait_present(self, entry_id: str, timeout: float = 120.0) -> bool:
+        """Block until the entry lands in the store (another task's settle
+        may be completing the transfer concurrently)."""
+        with self._cond:
+            return self._cond.wait_for(
+                lambda: entry_id in self.values, timeout=timeout
+            )
 
     def get(self, entry_id: str) -> Any:
         if self.spill.is_spilled(entry_id):
