@@ -122,8 +122,7 @@ def c4_source(seed: int) -> torch.Tensor:
     if _gpu():
         from lzy_amd.ops import fill_pattern
 
-        fill_pattern(t, seed=seed)
-        t.view(torch.int16).bitwise_and_(0x3FFF)
+        fill_pattern(t, seed=seed, mask16=0x3FFF)
     else:
         t.zero_()
     return t

@@ -268,3 +268,25 @@ def test_status_dashboard_html(storage_root):
     port = serve_status(storage_root=str(storage_root))
     html = urllib.request.urlopen(f"http://127.0.0.1:{port}/").read().decode()
     assert "<html" in html and "Whiteboards" in html and "/metrics" in html
+
+
+@pytest.mark.parametrize("cfg", [1, 2, 3, 4])
+def test_baseline_configs_cpu(cfg, tmp_path):
+    """BASELINE.json measurement configs stay runnable (CPU smoke; the
+    GPU numbers come from the driver/bench)."""
+    import json as _json
+    import subprocess
+    import sys
+    from pathlib import Path
+
+    root = Path(__file__).resolve().parent.parent
+    env = {**os.environ, "LZY_AMD_STORAGE": str(tmp_path / "s"),
+           "PYTHONPATH": str(root), "LZY_C4_GB": "0.02"}
+    r = subprocess.run(
+        [sys.executable, "benchmarks/baseline_configs.py", "--config",
+         str(cfg), "--iters", "1"],
+        cwd=root, env=env, capture_output=True, text=True, timeout=240,
+    )
+    assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
+    out = _json.loads(r.stdout.strip().splitlines()[-1])
+    assert out["config"] == cfg and out["makespan_s"] > 0
