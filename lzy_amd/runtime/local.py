@@ -52,6 +52,10 @@ def default_storage_uri() -> str:
 
 class LocalRuntime(Runtime):
     def __init__(self, max_workers: Optional[int] = None, journal_dir: Optional[str] = None):
+        if max_workers is None:
+            from lzy_amd.config import get_config
+
+            max_workers = get_config().dispatch_workers or None
         self._max_workers = max_workers or min(32, (os.cpu_count() or 8))
         self._journal_dir = journal_dir
         self._pool: Optional[ThreadPoolExecutor] = None
