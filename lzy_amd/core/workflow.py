@@ -105,7 +105,22 @@ class LzyWorkflow:
 
     # -- calls & barrier ------------------------------------------------------
 
+    def abort(self, reason: str = "aborted by user") -> None:
+        """Client-initiated abort (reference: AbortWorkflow/StopGraph —
+        LzyService.java:345, workflow.py:173): drop queued calls, abort
+        the runtime, and poison further use of this workflow."""
+        self._call_queue = []
+        self._aborted_reason = reason
+        try:
+            self.owner.runtime.abort(self)
+        finally:
+            raise WorkflowAbortedError(f"workflow {self.name}: {reason}")
+
     def register_call(self, call: "LzyCall") -> None:
+        if getattr(self, "_aborted_reason", None) is not None:
+            raise WorkflowAbortedError(
+                f"workflow {self.name} was aborted: {self._aborted_reason}"
+            )
         self._call_queue.append(call)
         if self.eager:
             self.barrier()

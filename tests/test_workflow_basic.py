@@ -226,3 +226,25 @@ def test_interactive_confirm(lzy, monkeypatch):
         raise AssertionError("expected WorkflowAbortedError")
     except WorkflowAbortedError:
         pass
+
+
+def test_client_abort(lzy):
+    """wf.abort() drops queued calls and surfaces WorkflowAbortedError
+    (reference: AbortWorkflow/StopGraph)."""
+    from lzy_amd.exceptions import WorkflowAbortedError
+
+    ran = []
+
+    @op
+    def record(x: int) -> int:
+        ran.append(x)
+        return x
+
+    try:
+        with lzy.workflow("abort-wf", interactive=False) as wf:
+            record(1)  # queued, never executed
+            wf.abort("changed my mind")
+        raise AssertionError("expected WorkflowAbortedError")
+    except WorkflowAbortedError as e:
+        assert "changed my mind" in str(e)
+    assert ran == []
