@@ -136,7 +136,12 @@ class WorkerAgent:
                 msg = self.ctrl.recv()
             except (EOFError, OSError):
                 break
-            self._handle(msg)
+            try:
+                self._handle(msg)
+            except BaseException as e:  # noqa: BLE001 - serve loop survives
+                _LOG.exception("agent r%d serve error on %s", self.rank,
+                               msg.get("cmd"))
+                self._report_error(msg, e)
 
     def _handle(self, msg: dict) -> None:
         cmd = msg["cmd"]

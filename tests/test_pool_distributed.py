@@ -113,3 +113,13 @@ def test_pool_world4_gangs_and_chunks(tmp_path):
     r = _run_distributed("tests/pool_script_world4.py", 4, tmp_path, timeout=240)
     assert r.returncode == 0, r.stdout[-3000:] + r.stderr[-3000:]
     assert "WORLD4-OK" in r.stdout
+
+
+def test_bench_dag_eight_ranks(tmp_path):
+    """Exactly the driver's 8-GPU SCALE shape, world 8 on gloo."""
+    r = _run_distributed(
+        "tests/bench_dag_script.py", 8, tmp_path,
+        extra_env={"LZY_BENCH_SHARD_MB": "1"}, timeout=300,
+    )
+    assert r.returncode == 0, r.stdout[-3000:] + r.stderr[-3000:]
+    assert "BENCH-DAG-OK" in r.stdout
