@@ -45,3 +45,26 @@ def gc_storage(
     if collect_whiteboards:
         _sweep("whiteboards", "whiteboards")
     return removed
+
+
+def gc_journals(
+    journal_dir: str,
+    ttl_seconds: float = 7 * 24 * 3600,
+    now: float | None = None,
+) -> int:
+    """Remove expired crash-resume journals (one per execution)."""
+    import os
+
+    now = now if now is not None else time.time()
+    base = Path(journal_dir)
+    n = 0
+    if not base.is_dir():
+        return 0
+    for child in base.glob("*.jsonl"):
+        try:
+            if now - child.stat().st_mtime > ttl_seconds:
+                os.unlink(child)
+                n += 1
+        except FileNotFoundError:
+            continue
+    return n

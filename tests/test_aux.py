@@ -290,3 +290,18 @@ def test_baseline_configs_cpu(cfg, tmp_path):
     assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
     out = _json.loads(r.stdout.strip().splitlines()[-1])
     assert out["config"] == cfg and out["makespan_s"] > 0
+
+
+def test_journal_gc(tmp_path):
+    import time as _t
+
+    from lzy_amd.storage.gc import gc_journals
+
+    d = tmp_path / "j"
+    d.mkdir()
+    (d / "old.jsonl").write_text("{}")
+    (d / "new.jsonl").write_text("{}")
+    old = _t.time() - 10_000
+    os.utime(d / "old.jsonl", (old, old))
+    assert gc_journals(str(d), ttl_seconds=3600) == 1
+    assert (d / "new.jsonl").exists() and not (d / "old.jsonl").exists()
