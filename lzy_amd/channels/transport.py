@@ -153,6 +153,11 @@ class Transport:
             and torch.cuda.is_available()
             and torch.cuda.device_count() >= world
         )
+        # config read once: chunk size is fixed for the transport's life
+        # (get_config() re-fingerprints the env — too hot per transfer)
+        from lzy_amd.config import get_config
+
+        self._chunk_bytes = get_config().channel_chunk_mb << 20
 
     # -- chunking ------------------------------------------------------------
     #
@@ -167,9 +172,7 @@ class Transport:
     # pairs them 1:1 with the receiver's chunk recvs.
 
     def _chunk_elems(self, elem_size: int) -> int:
-        from lzy_amd.config import get_config
-
-        return max(1, (get_config().channel_chunk_mb << 20) // max(1, elem_size))
+        return max(1, self._chunk_bytes // max(1, elem_size))
 
     @staticmethod
     def _chunks(flat: torch.Tensor, per: int, offset_chunks: int):

@@ -642,6 +642,9 @@ class _DriverScheduler:
         # one gang task inflight at a time, the rest queue here
         self.gang_inflight: Optional[str] = None
         self.deferred_gangs: List[str] = []
+        from lzy_amd.channels.transport import ipc_enabled
+
+        self._ipc_mode = ipc_enabled()  # config read once per batch
 
     # -- metadata helpers ---------------------------------------------------
 
@@ -779,10 +782,10 @@ class _DriverScheduler:
                     eid
                 ) or pickle_value(self.pool.agent.store.get(eid))
                 continue
-            from lzy_amd.channels.transport import KIND_TENSOR as _KT, ipc_enabled
+            from lzy_amd.channels.transport import KIND_TENSOR as _KT
 
             use_ipc = (
-                ipc_enabled()
+                self._ipc_mode
                 and meta.kind == _KT
                 and meta.device_type == "cuda"
             )
