@@ -57,14 +57,20 @@ class StorageRegistry:
         self._default: Optional[str] = None
 
     def register_storage(self, name: str, config: StorageConfig, default: bool = False) -> None:
-        from lzy_amd.storage.fs import FsStorageClient
+        if config.uri.startswith("file://"):
+            from lzy_amd.storage.fs import FsStorageClient
 
-        if not config.uri.startswith("file://"):
+            client: "StorageClient" = FsStorageClient()
+        elif config.uri.startswith("mem://"):
+            from lzy_amd.storage.mem import MemStorageClient
+
+            client = MemStorageClient()
+        else:
             raise ValueError(
                 f"Unsupported storage scheme for single-node runtime: {config.uri}"
             )
         self._configs[name] = config
-        self._clients[name] = FsStorageClient()
+        self._clients[name] = client
         if default or self._default is None:
             self._default = name
 

@@ -130,3 +130,22 @@ def test_user_serializer_transport(reg):
     other = LzySerializerRegistry()
     other.load_user_serializers(payload)
     assert other.find_serializer_by_data_format("token_raw") is not None
+
+
+def test_mem_storage_client():
+    """mem:// blobs (reference: InMemoryS3Storage test fleet)."""
+    from lzy_amd.storage.api import StorageConfig, StorageRegistry
+    from lzy_amd.storage.mem import reset_mem_storage
+
+    reset_mem_storage()
+    reg = StorageRegistry()
+    reg.register_storage("m", StorageConfig(uri="mem://t"), default=True)
+    c = reg.default_client()
+    c.write_bytes("mem://t/a/b", b"payload")
+    assert c.blob_exists("mem://t/a/b")
+    assert c.read_bytes("mem://t/a/b") == b"payload"
+    assert c.size_in_bytes("mem://t/a/b") == 7
+    c.copy("mem://t/a/b", "mem://t/c")
+    assert c.read_bytes("mem://t/c") == b"payload"
+    reset_mem_storage()
+    assert not c.blob_exists("mem://t/a/b")
