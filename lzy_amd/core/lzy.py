@@ -129,6 +129,17 @@ class Lzy(WithEnvironmentMixin):
             return None
         return ReadOnlyWhiteboard(meta, self)
 
+    def executions(self, limit: int = 100) -> Sequence[dict]:
+        """Recent executions with task-state counts from the crash-resume
+        journals (reference: the site/frontend task listing; served over
+        HTTP by utils/status.py — this is the in-process accessor)."""
+        import tempfile
+
+        from lzy_amd.utils.status import _workflows_payload
+
+        jdir = os.path.join(tempfile.gettempdir(), "lzy_amd_journal")
+        return _workflows_payload(jdir)[:limit]
+
     def whiteboards(
         self,
         *,

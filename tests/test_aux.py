@@ -305,3 +305,16 @@ def test_journal_gc(tmp_path):
     os.utime(d / "old.jsonl", (old, old))
     assert gc_journals(str(d), ttl_seconds=3600) == 1
     assert (d / "new.jsonl").exists() and not (d / "old.jsonl").exists()
+
+
+def test_lzy_executions_listing(lzy):
+    @op
+    def idf(x: int) -> int:
+        return x
+
+    with lzy.workflow("exec-list-wf"):
+        int(idf(1))
+    entries = lzy.executions()
+    assert any(e["execution_id"].startswith("exec-list-wf") for e in entries)
+    e = next(x for x in entries if x["execution_id"].startswith("exec-list-wf"))
+    assert e["tasks"] >= 1 and "done" in e["states"]
