@@ -69,3 +69,49 @@ def test_journal_written(tmp_path):
     states = Journal.replay(journals[-1])
     assert len(states) == 3
     assert all(s == "done" for s in states.values())
+
+
+def test_pool_crash_then_resume(tmp_path):
+    """Crash the whole pool mid-graph; a fresh pool resumes from the
+    result cache (GpuPoolRuntime durable-resume path)."""
+    import subprocess
+    import sys
+    from pathlib import Path
+
+    root = Path(__file__).resolve().parent.parent
+    marks = tmp_path / "marks"
+    marks.mkdir()
+
+    def run(crash: bool):
+        import socket
+
+        with socket.socket() as s:
+            s.bind(("127.0.0.1", 0))
+            port = s.getsockname()[1]
+        env = dict(os.environ)
+        env.update(
+            LZY_AMD_STORAGE=str(tmp_path / "storage"),
+            PYTHONPATH=str(root) + os.pathsep + env.get("PYTHONPATH", ""),
+            MARK_DIR=str(marks),
+            CRASH="1" if crash else "0",
+        )
+        env.pop("RANK", None)
+        env.pop("WORLD_SIZE", None)
+        return subprocess.run(
+            [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+             "--nproc-per-node=2", "--master-addr", "127.0.0.1",
+             "--master-port", str(port), "tests/pool_crash_script.py"],
+            cwd=root, env=env, capture_output=True, text=True, timeout=180,
+        )
+
+    r1 = run(crash=True)
+    assert r1.returncode != 0  # the hard exit propagates
+    def count(stage):
+        return sum(1 for f in marks.iterdir() if f.name.startswith(stage))
+    assert count("stage1") == 1 and count("stage2") == 1 and count("stage3") == 0
+
+    r2 = run(crash=False)
+    assert r2.returncode == 0, r2.stdout[-2000:] + r2.stderr[-2000:]
+    assert "RESULT=15" in r2.stdout and "POOL-RESUME-OK" in r2.stdout
+    # stage1/2 served from cache: no second execution
+    assert count("stage1") == 1 and count("stage2") == 1 and count("stage3") == 1
