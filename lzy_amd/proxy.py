@@ -108,6 +108,11 @@ def proxy(
     # proxy_adapter.py:28-33).
     ns["__lzy_materialized__"] = property(lambda self: state["done"])
     ns["__lzy_origin__"] = property(lambda self: __lzy_materialize__())
+    # isinstance(proxy, T) reflects the MATERIALIZED value's type —
+    # CPython's isinstance consults __class__ when the C-level type
+    # check fails (reference: test_simple_isinstance/.._none semantics:
+    # a None-valued Optional proxy is an instance of NoneType, not T).
+    ns["__class__"] = property(__instancecheck_getter)
 
     type_names = "_".join(getattr(t, "__name__", "obj") for t in types) or "obj"
     cls = type(f"LzyProxy_{type_names}", (), ns)

@@ -98,3 +98,49 @@ def test_hash_forwarding():
     p = proxy(lambda: "key", (str,))
     assert hash(p) == hash("key")
     assert {p: 1}[hash and "key"] == 1
+
+
+def _proxy_of(typ, value):
+    return proxy(lambda: value, (typ,))
+
+
+def test_iteration_and_unpacking():
+    p = _proxy_of(list, [1, 2, 3])
+    assert list(iter(p)) == [1, 2, 3]
+    a, b, c = p
+    assert (a, b, c) == (1, 2, 3)
+    assert 2 in p and 9 not in p
+
+
+def test_isinstance_of_declared_type():
+    p = _proxy_of(dict, {"k": 1})
+    assert isinstance(p, dict)
+    assert p["k"] == 1
+    assert len(p) == 1
+
+
+def test_context_manager_forwarding(tmp_path):
+    f = tmp_path / "x.txt"
+    f.write_text("hello")
+    p = _proxy_of(object, open(f))
+    with p as fh:
+        assert fh.read() == "hello"
+
+
+def test_format_and_repr():
+    p = _proxy_of(float, 3.5)
+    assert f"{p:.2f}" == "3.50"
+    assert "3.5" in repr(p)
+
+
+def test_callable_result():
+    p = _proxy_of(object, lambda x: x * 3)
+    assert p(4) == 12
+
+
+def test_isinstance_none_optional():
+    """Optional[T] proxy that materializes to None: isinstance follows
+    the value (reference test_simple_isinstance_none)."""
+    p = proxy(lambda: None, (int, type(None)))
+    assert not isinstance(p, int)
+    assert isinstance(p, type(None))
