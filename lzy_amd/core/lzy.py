@@ -37,8 +37,16 @@ def lzy_auth(*, user: str, key_path: str = "", endpoint: str = "",
 
 def _default_runtime() -> Runtime:
     # under torchrun (one process per GPU) the pool runtime is the engine;
-    # otherwise the in-process local runtime.
-    if os.environ.get("WORLD_SIZE") and int(os.environ.get("WORLD_SIZE", "1")) > 1:
+    # otherwise the in-process local runtime.  Inside an op body (pool
+    # worker thread) a nested Lzy() must NOT try to become a pool driver:
+    # nested workflows run in-process (scenarios/nested_workflows).
+    from lzy_amd.runtime.context import in_op_execution
+
+    if (
+        not in_op_execution()
+        and os.environ.get("WORLD_SIZE")
+        and int(os.environ.get("WORLD_SIZE", "1")) > 1
+    ):
         from lzy_amd.runtime.pool import GpuPoolRuntime
 
         return GpuPoolRuntime()

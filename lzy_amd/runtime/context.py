@@ -43,3 +43,19 @@ def _set_op_context(ctx: Optional[OpContext]) -> None:
 def op_context() -> Optional[OpContext]:
     """The gang context of the currently executing op (None outside gangs)."""
     return getattr(_ctx, "value", None)
+
+
+_in_task = threading.local()
+
+
+def _set_in_op_execution(flag: bool) -> None:
+    _in_task.value = flag
+
+
+def in_op_execution() -> bool:
+    """True on a thread currently executing an op body.  Nested
+    ``Lzy().workflow()`` started from inside an op must run in-process
+    (LocalRuntime) — a pool worker cannot become a pool driver
+    (reference analogue: nested graphs run through the op's own client,
+    scenarios/nested_workflows)."""
+    return bool(getattr(_in_task, "value", False))

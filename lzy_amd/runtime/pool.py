@@ -941,16 +941,29 @@ class _DriverScheduler:
         self.outstanding[rank] -= 1
         call = self.calls[result.task_id]
         gang = self.gang_pending.get(result.task_id)
-        exc_name, exc_msg, tb = unpickle_value(result.exc_bytes)
+        payload = unpickle_value(result.exc_bytes)
+        exc_name, exc_msg, tb = payload[0], payload[1], payload[2]
+        exc_blob = payload[3] if len(payload) > 3 else None
         self.workflow.snapshot.put(call.exception_id, (exc_name, exc_msg, tb))
-        self.errors.append(
-            LzyExecutionError(
-                f"Op {call.callable_name} failed on rank {rank}: "
-                f"{exc_name}: {exc_msg}",
-                task_id=result.task_id,
-                remote_traceback=tb,
-            )
+        err = LzyExecutionError(
+            f"Op {call.callable_name} failed on rank {rank}: "
+            f"{exc_name}: {exc_msg}",
+            task_id=result.task_id,
+            remote_traceback=tb,
         )
+        if exc_blob is not None:
+            try:
+                # re-raise chain carries the user's exception object,
+                # rebuilt without calling a possibly-incompatible __init__
+                cls, args, state = unpickle_value(exc_blob)
+                cause = cls.__new__(cls)
+                cause.args = args
+                if state:
+                    cause.__dict__.update(state)
+                err.__cause__ = cause
+            except Exception:  # noqa: BLE001 - string triple remains
+                pass
+        self.errors.append(err)
         self.journal.record(result.task_id, "failed", exc_msg)
         if gang is not None:
             gang.discard(rank)

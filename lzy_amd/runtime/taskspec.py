@@ -198,7 +198,10 @@ def run_taskspec(
 
         from lzy_amd.runtime.streams import STREAMS
 
+        from lzy_amd.runtime.context import _set_in_op_execution
+
         stream = STREAMS.next_stream()
+        _set_in_op_execution(True)
         with (torch.cuda.stream(stream) if stream is not None else nullcontext()):
             for eid, v in zip(
                 list(spec.arg_entries) + list(spec.kwarg_entries.values()),
@@ -209,6 +212,7 @@ def run_taskspec(
     except BaseException as e:  # noqa: BLE001 - transported as a value
         return _fail(spec, t0, e, traceback.format_exc())
     finally:
+        _set_in_op_execution(False)
         if spec.gang is not None:
             from lzy_amd.runtime.context import _set_op_context
 
@@ -300,7 +304,20 @@ def _fail(spec: TaskSpec, t0: float, exc: BaseException, tb: str) -> TaskResult:
     from lzy_amd.channels.transport import pickle_value
 
     METRICS.inc("lzy_op_failures", op=spec.name)
-    payload = pickle_value((type(exc).__name__, str(exc), tb))
+    # ship the exception OBJECT too (reference: pickled exc_info written
+    # as a first-class output, startup.py:124-169) so the client re-raises
+    # the user's exception type with its attributes.  Packed as
+    # (class, args, __dict__) and rebuilt with __new__ — default pickling
+    # breaks on exception classes whose __init__ signature differs from
+    # their .args.  Falls back to the string triple when unpicklable.
+    exc_blob: Optional[bytes] = None
+    try:
+        exc_blob = pickle_value(
+            (type(exc), tuple(exc.args), dict(getattr(exc, "__dict__", {})))
+        )
+    except Exception:  # noqa: BLE001
+        pass
+    payload = pickle_value((type(exc).__name__, str(exc), tb, exc_blob))
     return TaskResult(
         task_id=spec.task_id, ok=False, exc_bytes=payload,
         elapsed_s=time.perf_counter() - t0,

@@ -80,3 +80,48 @@ def test_scenario(name, tmp_path):
     # expectations read like plain program output
     actual = [re.sub(r"^\[LZY-[^\]]*\] ", "", l) for l in r.stdout.splitlines()]
     _match(expected, actual, name)
+
+
+# the same scenario scripts under the pool engine (torchrun world 2):
+# "one script, both runtimes" — GpuPoolRuntime is picked from the launch
+# context, ops run on worker ranks, values cross the gloo/RCCL data plane
+_POOL_SCENARIOS = [
+    "complex_graph", "repeated_ops_use_cache", "fully_cached_graph",
+    "exec_fail", "cached_exception", "custom_serializer", "file_test",
+    "nested_workflows", "whiteboards", "two_execution_one_wf",
+    "exception_serialize",
+]
+
+
+@pytest.mark.parametrize("name", [n for n in _POOL_SCENARIOS if n in _names])
+def test_scenario_pool(name, tmp_path):
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    env = dict(os.environ)
+    env["LZY_AMD_STORAGE"] = str(tmp_path / "storage")
+    env["PYTHONPATH"] = str(ROOT) + os.pathsep + env.get("PYTHONPATH", "")
+    env.pop("RANK", None)
+    env.pop("WORLD_SIZE", None)
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node=2", "--master-addr", "127.0.0.1",
+         "--master-port", str(port), str(SCEN / name / "__init__.py")],
+        cwd=ROOT, env=env, capture_output=True, text=True, timeout=300,
+    )
+    assert r.returncode == 0, (
+        f"[{name}/pool] rc={r.returncode}\n--- stdout ---\n{r.stdout[-4000:]}"
+        f"\n--- stderr ---\n{r.stderr[-4000:]}"
+    )
+    expected = (SCEN / name / "expected_stdout").read_text().splitlines()
+    actual = [
+        re.sub(r"^\[LZY-[^\]]*\] ", "", l)
+        for l in r.stdout.splitlines()
+        # torchrun/gloo startup noise is not scenario output
+        if l.strip()
+        and not l.startswith("[Gloo]") and " is connected to " not in l
+        and not re.match(r"^[WIE]\d{4}", l)
+    ]
+    _match(expected, actual, f"{name}/pool")
