@@ -330,7 +330,12 @@ class WorkerAgent:
             tag = spec.gang["tag"]
             gang_group = self._groups.get(tag)  # None -> default pg
         result = run_taskspec(
-            spec, self.store, self.serializers, self.storage, gang_group=gang_group
+            spec, self.store, self.serializers, self.storage,
+            gang_group=gang_group,
+            # rank 0 echoes op logs on its own console; other ranks ship
+            # them in the TaskResult and the driver tails them (the
+            # console belongs to the client)
+            echo_logs=(self.rank == 0),
         )
         self.ctrl.send_event(
             {
@@ -895,9 +900,23 @@ class _DriverScheduler:
 
     # -- completion ----------------------------------------------------------
 
+    def _tail_logs(self, rank: int, result: TaskResult) -> None:
+        """Print a remote rank's captured op logs on the client console
+        (reference: ReadStdSlots live tail, runtime.py:283-301)."""
+        if rank == 0:
+            return  # echoed live on the shared console already
+        name = self.calls[result.task_id].callable_name
+        if result.logs_out:
+            for line in result.logs_out.splitlines():
+                print(f"[LZY-{name}] {line}", flush=True)
+        if result.logs_err:
+            for line in result.logs_err.splitlines():
+                print(f"[LZY-{name}] {line}", file=sys.stderr, flush=True)
+
     def _on_done(self, rank: int, result: TaskResult) -> bool:
         """Returns True when the task fully completed (all gang members)."""
         self.outstanding[rank] -= 1
+        self._tail_logs(rank, result)
         call = self.calls[result.task_id]
         gang = self.gang_pending.get(result.task_id)
         primary = self.task_ranks[result.task_id][0]
@@ -939,6 +958,7 @@ class _DriverScheduler:
 
     def _on_failed(self, rank: int, result: TaskResult) -> None:
         self.outstanding[rank] -= 1
+        self._tail_logs(rank, result)
         call = self.calls[result.task_id]
         gang = self.gang_pending.get(result.task_id)
         payload = unpickle_value(result.exc_bytes)

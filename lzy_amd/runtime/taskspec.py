@@ -50,6 +50,10 @@ class TaskResult:
     exc_bytes: Optional[bytes] = None
     elapsed_s: float = 0.0
     cached: bool = False
+    # captured op std-logs, tailed on the driver's console (reference:
+    # worker -> Kafka -> client ReadStdSlots stream); capped at 256 KiB
+    logs_out: str = ""
+    logs_err: str = ""
 
 
 class WorkerStore:
@@ -115,12 +119,16 @@ class WorkerStore:
         self.pickled.clear()
 
 
+_LOG_CAP = 256 << 10
+
+
 def run_taskspec(
     spec: TaskSpec,
     store: WorkerStore,
     serializers,
     storage,
     gang_group=None,
+    echo_logs: bool = True,
 ) -> TaskResult:
     """Execute one TaskSpec against the worker store (worker innermost loop)."""
     from lzy_amd.channels.transport import describe_value, pickle_value, unpickle_value
@@ -171,7 +179,7 @@ def run_taskspec(
     func = _load_func(spec.func_bytes)
 
     capture = OpLogCapture.instance()
-    out_buf, err_buf = capture.route_current_thread(spec.name)
+    out_buf, err_buf = capture.route_current_thread(spec.name, echo=echo_logs)
     old_env: Dict[str, Optional[str]] = {}
     gang_env: Dict[str, str] = {}
     if spec.gang is not None:
@@ -210,7 +218,10 @@ def run_taskspec(
                 STREAMS.wait_value(eid, v)
             result = func(*args, **kwargs)
     except BaseException as e:  # noqa: BLE001 - transported as a value
-        return _fail(spec, t0, e, traceback.format_exc())
+        tr = _fail(spec, t0, e, traceback.format_exc())
+        tr.logs_out = out_buf.getvalue()[:_LOG_CAP]
+        tr.logs_err = err_buf.getvalue()[:_LOG_CAP]
+        return tr
     finally:
         _set_in_op_execution(False)
         if spec.gang is not None:
@@ -264,7 +275,11 @@ def run_taskspec(
     elapsed = time.perf_counter() - t0
     METRICS.observe("lzy_op_run", elapsed)
     METRICS.observe(f"lzy_op::{spec.name}", elapsed)
-    return TaskResult(task_id=spec.task_id, ok=True, outputs=outputs, elapsed_s=elapsed)
+    return TaskResult(
+        task_id=spec.task_id, ok=True, outputs=outputs, elapsed_s=elapsed,
+        logs_out=out_buf.getvalue()[:_LOG_CAP],
+        logs_err=err_buf.getvalue()[:_LOG_CAP],
+    )
 
 
 def _archive_logs(spec: TaskSpec, storage, out: str, err: str) -> None:

@@ -22,10 +22,10 @@ class _ThreadRoutingTee(io.TextIOBase):
         self._routes: Dict[int, Tuple[str, io.StringIO]] = {}
         self._lock = threading.Lock()
 
-    def route(self, prefix: str) -> io.StringIO:
+    def route(self, prefix: str, echo: bool = True) -> io.StringIO:
         buf = io.StringIO()
         with self._lock:
-            self._routes[threading.get_ident()] = (prefix, buf)
+            self._routes[threading.get_ident()] = (prefix, buf, echo)
         return buf
 
     def unroute(self) -> None:
@@ -38,10 +38,11 @@ class _ThreadRoutingTee(io.TextIOBase):
             entry = self._routes.get(ident)
         if entry is None:
             return self._fallback.write(s)
-        prefix, buf = entry
+        prefix, buf, echo = entry
         buf.write(s)
-        for line in s.splitlines(keepends=True):
-            self._fallback.write(f"{prefix}{line}" if line.strip() else line)
+        if echo:
+            for line in s.splitlines(keepends=True):
+                self._fallback.write(f"{prefix}{line}" if line.strip() else line)
         return len(s)
 
     def flush(self) -> None:
@@ -91,10 +92,16 @@ class OpLogCapture:
                 self._out_tee = None
                 self._err_tee = None
 
-    def route_current_thread(self, task_name: str) -> Tuple[io.StringIO, io.StringIO]:
+    def route_current_thread(
+        self, task_name: str, echo: bool = True
+    ) -> Tuple[io.StringIO, io.StringIO]:
+        """``echo=False``: capture only — pool workers on ranks > 0 do not
+        write to their own console; their logs travel to the driver in the
+        TaskResult and are tailed there (reference: worker→Kafka→client
+        ReadStdSlots stream)."""
         prefix = f"[LZY-{task_name}] "
-        out = self._out_tee.route(prefix) if self._out_tee else io.StringIO()
-        err = self._err_tee.route(prefix) if self._err_tee else io.StringIO()
+        out = self._out_tee.route(prefix, echo) if self._out_tee else io.StringIO()
+        err = self._err_tee.route(prefix, echo) if self._err_tee else io.StringIO()
         return out, err
 
     def unroute_current_thread(self) -> None:

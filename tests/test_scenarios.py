@@ -105,21 +105,32 @@ def test_scenario_pool(name, tmp_path):
     env["PYTHONPATH"] = str(ROOT) + os.pathsep + env.get("PYTHONPATH", "")
     env.pop("RANK", None)
     env.pop("WORLD_SIZE", None)
+    log_dir = tmp_path / "ranklogs"
+    log_dir.mkdir()
     r = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node=2", "--master-addr", "127.0.0.1",
-         "--master-port", str(port), str(SCEN / name / "__init__.py")],
+         "--master-port", str(port),
+         # per-rank output files: cross-rank interleaving would split
+         # lines mid-write on a shared console
+         "--log-dir", str(log_dir), "--redirects", "3",
+         str(SCEN / name / "__init__.py")],
         cwd=ROOT, env=env, capture_output=True, text=True, timeout=300,
     )
+    rank0_out = ""
+    for f in sorted(log_dir.rglob("stdout*")):
+        if f.parent.name == "0" or "/0/" in str(f):
+            rank0_out = f.read_text()
+            break
     assert r.returncode == 0, (
-        f"[{name}/pool] rc={r.returncode}\n--- stdout ---\n{r.stdout[-4000:]}"
-        f"\n--- stderr ---\n{r.stderr[-4000:]}"
+        f"[{name}/pool] rc={r.returncode}\n--- rank0 stdout ---\n"
+        f"{rank0_out[-4000:]}\n--- stderr ---\n{r.stderr[-4000:]}"
     )
     expected = (SCEN / name / "expected_stdout").read_text().splitlines()
     actual = [
         re.sub(r"^\[LZY-[^\]]*\] ", "", l)
-        for l in r.stdout.splitlines()
-        # torchrun/gloo startup noise is not scenario output
+        for l in rank0_out.splitlines()
+        # gloo prints its connection banner on stdout at pg init
         if l.strip()
         and not l.startswith("[Gloo]") and " is connected to " not in l
         and not re.match(r"^[WIE]\d{4}", l)
