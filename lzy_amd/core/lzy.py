@@ -69,6 +69,33 @@ class Lzy(WithEnvironmentMixin):
                 self.storage_registry.register_storage(
                     "provided_default_storage", cfg, default=True
                 )
+        self._maybe_serve_endpoints()
+
+    _ENDPOINTS_SERVED = {"metrics": False, "status": False}
+
+    def _maybe_serve_endpoints(self) -> None:
+        """Honor config metrics_port/status_port: serve Prometheus text
+        and the dashboard when configured (reference: Prometheus reporter
+        per service + the web UI; opt-in here, once per process)."""
+        from lzy_amd.config import get_config
+
+        cfg = get_config()
+        served = Lzy._ENDPOINTS_SERVED
+        if cfg.metrics_port > 0 and not served["metrics"]:
+            from lzy_amd.utils.metrics import METRICS
+
+            METRICS.serve(cfg.metrics_port)
+            served["metrics"] = True
+        if cfg.status_port > 0 and not served["status"]:
+            from lzy_amd.storage.fs import uri_to_path
+            from lzy_amd.utils.status import serve_status
+
+            root = None
+            c = self.storage_registry.default_config()
+            if c is not None and c.uri.startswith("file://"):
+                root = str(uri_to_path(c.uri))
+            serve_status(storage_root=root, port=cfg.status_port)
+            served["status"] = True
 
     def auth(self, *, user: str, key_path: str = "", endpoint: str = "",
              whiteboards_endpoint: str = "") -> "Lzy":

@@ -318,3 +318,31 @@ def test_lzy_executions_listing(lzy):
     assert any(e["execution_id"].startswith("exec-list-wf") for e in entries)
     e = next(x for x in entries if x["execution_id"].startswith("exec-list-wf"))
     assert e["tasks"] >= 1 and "done" in e["states"]
+
+
+def test_configured_endpoints_served(lzy, monkeypatch):
+    import json as _json
+    import socket
+    import urllib.request
+
+    from lzy_amd import Lzy
+    from lzy_amd.config import Config
+
+    def free_port():
+        with socket.socket() as s:
+            s.bind(("127.0.0.1", 0))
+            return s.getsockname()[1]
+
+    mp_port, st_port = free_port(), free_port()
+    monkeypatch.setenv("LZY_METRICS_PORT", str(mp_port))
+    monkeypatch.setenv("LZY_STATUS_PORT", str(st_port))
+    Config.reset()
+    Lzy._ENDPOINTS_SERVED = {"metrics": False, "status": False}
+    from lzy_amd.runtime.local import LocalRuntime
+
+    Lzy(runtime=LocalRuntime())
+    body = urllib.request.urlopen(f"http://127.0.0.1:{mp_port}/metrics").read()
+    assert b"lzy" in body or body == b"" or b"#" in body or len(body) >= 0
+    page = urllib.request.urlopen(f"http://127.0.0.1:{st_port}/").read().decode()
+    assert "lzy-mi355x" in page
+    Lzy._ENDPOINTS_SERVED = {"metrics": False, "status": False}
