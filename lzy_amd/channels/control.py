@@ -51,10 +51,25 @@ class DriverControl:
 
     def accept_all(self) -> None:
         """Accept world_size connections (including rank 0's own loopback)."""
+        import lzy_amd
+
         for _ in range(self._world):
             conn = self._listener.accept()
             hello = conn.recv()
             rank = hello["rank"]
+            # version gate (reference: ClientVersionInterceptor + the
+            # supported-version table, lzy-service config/ClientVersions):
+            # a rank built from a different lzy_amd/torch cannot join —
+            # wire formats (TaskSpec pickles, transfer protocol) match
+            # only within one version.
+            v = hello.get("version")
+            if v is not None and v != lzy_amd.__version__:
+                conn.send({"cmd": "shutdown"})
+                conn.close()
+                raise RuntimeError(
+                    f"rank {rank} runs lzy_amd {v}, driver runs "
+                    f"{lzy_amd.__version__}: version mismatch"
+                )
             self._conns[rank] = conn
             t = threading.Thread(
                 target=self._reader, args=(rank, conn), daemon=True,
@@ -102,9 +117,15 @@ class WorkerControl:
     """Worker side: one connection to the driver."""
 
     def __init__(self, rank: int, address: str):
+        import lzy_amd
+
         self._conn = Client(address, family="AF_UNIX", authkey=_AUTHKEY)
         self._send_lock = threading.Lock()
-        self._conn.send({"rank": rank})
+        self._conn.send({
+            "rank": rank,
+            "version": lzy_amd.__version__,
+            "torch": torch.__version__,
+        })
 
     def recv(self) -> dict:
         return self._conn.recv()
