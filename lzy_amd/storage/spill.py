@@ -105,6 +105,11 @@ class SpillManager:
         self._lock = threading.Lock()
         self._lru: "OrderedDict[str, int]" = OrderedDict()  # entry -> nbytes
         self._spilled: Dict[str, Tuple[torch.Tensor, torch.device, Any]] = {}
+        self._capacity_cache: Optional[int] = None
+        # allowance for HBM the store does not track (model weights,
+        # activations): the real allocator is consulted once tracked
+        # bytes approach the threshold
+        self._untracked_slack = 32 << 30
 
     # -- bookkeeping --------------------------------------------------------
 
@@ -137,7 +142,16 @@ class SpillManager:
         Returns the number of entries spilled."""
         if not self.enabled:
             return 0
-        limit = int(self._capacity() * self.threshold_frac)
+        # fast path: querying the real allocator (torch memory_stats walks
+        # the whole stats tree, ~0.1 ms) is pointless while the store's own
+        # tracked bytes are far below the threshold
+        if self._capacity_cache is None:
+            self._capacity_cache = self._capacity()
+        limit = int(self._capacity_cache * self.threshold_frac)
+        with self._lock:
+            tracked = sum(self._lru.values())
+        if tracked < limit // 2 and tracked + self._untracked_slack < limit:
+            return 0
         n = 0
         while self._bytes_in_use() > limit:
             with self._lock:
