@@ -51,6 +51,7 @@ class Config:
     # scheduler / pool
     dispatch_workers: int = 0          # 0 -> auto (LocalRuntime thread pool size)
     exec_threads: int = 2              # executor threads per pool rank
+    chain_dispatch: bool = True        # eager same-rank dependency dispatch
     gang_timeout_s: float = 120.0      # gang-allocation wait bound
     heartbeat_period_s: float = 2.0    # worker liveness probe period
     # result cache / snapshot

@@ -203,6 +203,9 @@ class WorkerAgent:
         elif cmd == "load_serializers":
             self.serializers.load_user_serializers(msg["payload"])
             self.ctrl.send_event({"ev": "ack", "tag": msg["tag"], "rank": self.rank})
+        elif cmd == "poison":
+            for eid in msg["entries"]:
+                self.store.poison(eid, msg.get("reason", "producer failed"))
         elif cmd == "drop_entries":
             for eid in msg["entries"]:
                 self.store.drop(eid)
@@ -647,6 +650,19 @@ class _DriverScheduler:
         # one gang task inflight at a time, the rest queue here
         self.gang_inflight: Optional[str] = None
         self.deferred_gangs: List[str] = []
+        # chain dispatch: a task whose producers are all inflight on ONE
+        # rank is dispatched there immediately (worker FIFO + store
+        # condition waits give correct ordering) — the driver round-trip
+        # leaves the dependency-chain critical path
+        self.task_deps: Dict[str, List[str]] = {}
+        self.entry_producer: Dict[str, str] = {}
+        self.dispatched: Set[str] = set()
+        self.dag_completed: Set[str] = set()
+        self.done_pending: Dict[str, str] = {}  # tid -> completed-not-dag-acked
+        self.chained_waits: Dict[str, Tuple[int, List[str]]] = {}
+        from lzy_amd.config import get_config
+
+        self._chain_enabled = bool(getattr(get_config(), "chain_dispatch", True))
         from lzy_amd.channels.transport import ipc_enabled
 
         # explicit ipc mode, or automatic when ranks outnumber GPUs:
@@ -692,11 +708,14 @@ class _DriverScheduler:
                 if eid in producer and producer[eid] != c.id
             })
             dag.add_task(c.id, deps)
+            self.task_deps[c.id] = deps
+        self.entry_producer = producer
         dag.seal()
         METRICS.observe("lzy_graph_build", time.perf_counter() - t0)
 
         for tid in dag.take_ready():
             self._dispatch(tid)
+        self._try_chain()
 
         pool = self.pool
         failed_tasks: Set[str] = set()
@@ -707,15 +726,36 @@ class _DriverScheduler:
                 result: TaskResult = msg["result"]
                 finished = self._on_done(rank, result)
                 if finished and result.task_id not in failed_tasks:
-                    for tid in dag.complete(result.task_id):
-                        self._dispatch(tid)
+                    # chained completions can outrace their parents'
+                    # events; feed the DAG in dependency order
+                    self.done_pending[result.task_id] = result.task_id
+                    progressed = True
+                    while progressed:
+                        progressed = False
+                        for t in list(self.done_pending):
+                            if all(d in self.dag_completed
+                                   for d in self.task_deps.get(t, ())):
+                                self.done_pending.pop(t)
+                                self.dag_completed.add(t)
+                                for nxt in dag.complete(t):
+                                    if nxt not in self.dispatched:
+                                        self._dispatch(nxt)
+                                progressed = True
+                    self._try_chain()
             elif ev == "task_failed":
                 result = msg["result"]
                 self._on_failed(rank, result)
                 if result.task_id not in failed_tasks:
                     failed_tasks.add(result.task_id)
                     for ct in dag.fail(result.task_id):
-                        self.journal.record(ct, "cancelled")
+                        if ct in self.dispatched:
+                            # chain-dispatched dependent already inflight:
+                            # poison its pending inputs so its settle
+                            # fails now instead of timing out
+                            failed_tasks.add(ct)
+                            self._poison_chained(ct, result.task_id)
+                        else:
+                            self.journal.record(ct, "cancelled")
             elif ev == "agent_error":
                 self.errors.append(
                     LzyExecutionError(f"agent rank {rank}: {msg['error']}")
@@ -748,7 +788,8 @@ class _DriverScheduler:
 
     # -- dispatch ------------------------------------------------------------
 
-    def _dispatch(self, task_id: str) -> None:
+    def _dispatch(self, task_id: str, chain_rank: Optional[int] = None,
+                  pending_local: Optional[Set[str]] = None) -> None:
         call = self.calls[task_id]
         gpu_count = call.env.provisioning.effective_gpu_count
         pool = self.pool
@@ -761,7 +802,10 @@ class _DriverScheduler:
             return
 
         dispatch_t0 = time.perf_counter()
-        if gpu_count > 1:
+        if chain_rank is not None:
+            ranks = [chain_rank]
+            gang = None
+        elif gpu_count > 1:
             if self.gang_inflight is not None:
                 self.deferred_gangs.append(task_id)
                 return
@@ -773,6 +817,7 @@ class _DriverScheduler:
             ranks = [self._pick_rank(call)]
             gang = None
 
+        self.dispatched.add(task_id)
         self.task_ranks[task_id] = ranks
         if gang is not None:
             self.gang_pending[task_id] = set(ranks)
@@ -783,6 +828,11 @@ class _DriverScheduler:
         recvs_by_rank: Dict[int, List[dict]] = {}
 
         for eid in call.input_entry_ids():
+            if pending_local and eid in pending_local:
+                # produced by an inflight task on this same rank: no
+                # transfer, the worker's settle waits on the store
+                wait_entries_per_rank[ranks[0]].append(eid)
+                continue
             meta = self.meta.get(eid)
             if meta is None:
                 # driver-captured arg (or earlier-batch result on rank 0)
@@ -874,6 +924,8 @@ class _DriverScheduler:
             pool.driver_ctrl.send(r, {"cmd": "task", "spec": spec})
             self.outstanding[r] += 1
         self.inflight += 1
+        if chain_rank is not None and pending_local:
+            self.chained_waits[task_id] = (chain_rank, sorted(pending_local))
         self.task_dispatch_ts[task_id] = dispatch_t0
         self.journal.record(task_id, "scheduled", call.callable_name)
         METRICS.observe("lzy_dispatch", time.perf_counter() - dispatch_t0)
@@ -938,6 +990,7 @@ class _DriverScheduler:
             self.gang_pending.pop(result.task_id, None)
             self._release_gang(result.task_id)
         self.inflight -= 1
+        self.chained_waits.pop(result.task_id, None)
         self.journal.record(result.task_id, "done")
         ts = self.task_dispatch_ts.pop(result.task_id, None)
         if ts is not None:
@@ -949,6 +1002,56 @@ class _DriverScheduler:
         if result.cached:
             METRICS.inc("lzy_cache_hits_pool")
         return True
+
+    def _poison_chained(self, child_tid: str, failed_tid: str) -> None:
+        hit = self.chained_waits.get(child_tid)
+        if hit is None:
+            return
+        rank, eids = hit
+        name = self.calls[failed_tid].callable_name
+        self.pool.driver_ctrl.send(rank, {
+            "cmd": "poison", "entries": eids,
+            "reason": f"producer op {name} failed",
+        })
+
+    def _try_chain(self) -> None:
+        """Dispatch tasks whose producers are all inflight/complete on one
+        rank, without waiting for their completion events."""
+        if not self._chain_enabled:
+            return
+        for tid, call in self.calls.items():
+            if tid in self.dispatched:
+                continue
+            if call.env.provisioning.effective_gpu_count > 1:
+                continue  # gangs stay on the completion-driven path
+            deps = self.task_deps.get(tid, [])
+            if not deps or any(d not in self.dispatched for d in deps):
+                continue
+            ranks = set()
+            inflight_deps = []
+            ok = True
+            for d in deps:
+                dr = self.task_ranks.get(d)
+                if dr is None or len(dr) != 1 or d in (
+                    self.gang_pending
+                ):
+                    ok = False
+                    break
+                ranks.add(dr[0])
+                if d not in self.dag_completed and d not in self.done_pending:
+                    inflight_deps.append(d)
+            if not ok or len(ranks) != 1 or not inflight_deps:
+                continue  # multi-rank deps (or all complete): normal path
+            r = next(iter(ranks))
+            if r not in self.outstanding:
+                continue  # rank excluded (death)
+            pending_local = {
+                eid
+                for eid in call.input_entry_ids()
+                if self.entry_producer.get(eid) in inflight_deps
+            }
+            self._dispatch(tid, chain_rank=r, pending_local=pending_local)
+            METRICS.inc("lzy_chain_dispatches")
 
     def _release_gang(self, task_id: str) -> None:
         if self.gang_inflight == task_id:

@@ -56,6 +56,14 @@ class TaskResult:
     logs_err: str = ""
 
 
+class _PoisonEntry:
+    """Marks an entry whose producer failed: chained consumers waiting on
+    it must fail promptly instead of timing out."""
+
+    def __init__(self, reason: str):
+        self.reason = reason
+
+
 class WorkerStore:
     """Per-rank value store: entry_id -> live python object (device tensors
     stay in HBM).  Non-tensor values keep a pre-pickled byte image so
@@ -92,11 +100,21 @@ class WorkerStore:
                 lambda: entry_id in self.values, timeout=timeout
             )
 
+    def poison(self, entry_id: str, reason: str) -> None:
+        """Producer failed: wake and fail any chained consumer waiting on
+        this entry."""
+        with self._cond:
+            if entry_id not in self.values:
+                self.values[entry_id] = _PoisonEntry(reason)
+                self._cond.notify_all()
+
     def get(self, entry_id: str) -> Any:
         if self.spill.is_spilled(entry_id):
             value = self.spill.unspill(entry_id, self.values)
             return value
         value = self.values[entry_id]
+        if isinstance(value, _PoisonEntry):
+            raise RuntimeError(f"input {entry_id} unavailable: {value.reason}")
         self.spill.track(entry_id, value)  # LRU touch
         from lzy_amd.runtime.streams import STREAMS
 

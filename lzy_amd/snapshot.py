@@ -194,6 +194,11 @@ class DefaultSnapshot:
         (pool workers included) can load the blob without type context.
         """
         entry = self._entries[entry_id]
+        if entry_id not in self._values:
+            # hot-tier miss: the value may live on a remote owner rank
+            got = self.try_get(entry_id)  # triggers the fetcher
+            if not got.found:
+                raise KeyError(f"No value for entry {entry_id} to persist")
         value = self._values[entry_id]
         from lzy_amd.runtime.streams import STREAMS
 

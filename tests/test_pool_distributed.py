@@ -123,3 +123,9 @@ def test_bench_dag_eight_ranks(tmp_path):
     )
     assert r.returncode == 0, r.stdout[-3000:] + r.stderr[-3000:]
     assert "BENCH-DAG-OK" in r.stdout
+
+
+def test_chain_dispatch(tmp_path):
+    r = _run_distributed("tests/pool_script_chain.py", 2, tmp_path, timeout=240)
+    assert r.returncode == 0, r.stdout[-3000:] + r.stderr[-3000:]
+    assert "CHAIN-OK" in r.stdout
