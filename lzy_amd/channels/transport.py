@@ -134,6 +134,66 @@ def ipc_enabled() -> bool:
     return get_config().channel_transport == "ipc"
 
 
+# ---------------------------------------------------------------------------
+# cast-on-the-wire (BASELINE north star: the slot serializer's pack/unpack
+# + dtype-cast as a CDNA4 HIP kernel).  Opt-in (lossy): float tensors are
+# cast to a narrower wire dtype by the producer's cast kernel, halving (or
+# quartering) the bytes crossing xGMI, and cast back on the consumer.
+# Both sides derive the decision from (meta.dtype, config) — deterministic.
+# ---------------------------------------------------------------------------
+
+_WIRE_DTYPES = {
+    "fp16": torch.float16,
+    "bf16": torch.bfloat16,
+    "fp8e4m3": torch.float8_e4m3fn,
+    "fp8e5m2": torch.float8_e5m2,
+}
+_WIRE_CASTABLE = (torch.float32, torch.bfloat16, torch.float16)
+_WIRE_MIN_ELEMS = 1 << 16
+
+
+def wire_cast_dtype() -> Optional[torch.dtype]:
+    from lzy_amd.config import get_config
+
+    return _WIRE_DTYPES.get(get_config().channel_wire_cast)
+
+
+def _should_wirecast(dtype: torch.dtype, numel: int, wire) -> bool:
+    return (
+        wire is not None
+        and dtype in _WIRE_CASTABLE
+        and dtype != wire
+        and torch.empty(0, dtype=wire).element_size()
+        < torch.empty(0, dtype=dtype).element_size()
+        and numel >= _WIRE_MIN_ELEMS
+    )
+
+
+def wire_pack(t: torch.Tensor, wire: torch.dtype) -> torch.Tensor:
+    """Producer-side pack: cast to the wire dtype (HIP cast kernel on
+    device, torch cast on host)."""
+    out = torch.empty(t.shape, dtype=wire, device=t.device)
+    if t.is_cuda:
+        from lzy_amd import ops
+
+        if ops.NATIVE:
+            ops.cast_copy(t, out)
+            return out
+    return out.copy_(t.to(wire))
+
+
+def wire_unpack(w: torch.Tensor, dtype: torch.dtype) -> torch.Tensor:
+    """Consumer-side unpack: cast back to the declared dtype."""
+    out = torch.empty(w.shape, dtype=dtype, device=w.device)
+    if w.is_cuda:
+        from lzy_amd import ops
+
+        if ops.NATIVE:
+            ops.cast_copy(w, out)
+            return out
+    return out.copy_(w.to(dtype))
+
+
 class Transport:
     """Per-rank transfer engine over a dedicated data process group.
 
@@ -158,6 +218,7 @@ class Transport:
         from lzy_amd.config import get_config
 
         self._chunk_bytes = get_config().channel_chunk_mb << 20
+        self._wire = wire_cast_dtype()
 
     # -- chunking ------------------------------------------------------------
     #
@@ -199,6 +260,8 @@ class Transport:
             t = value.detach()
             if not t.is_contiguous():
                 t = t.contiguous()
+            if _should_wirecast(t.dtype, t.numel(), self._wire):
+                t = wire_pack(t, self._wire)  # cast kernel on device
             if t.is_cuda and not self._cuda_p2p:
                 t = t.cpu()
             per = self._chunk_elems(t.element_size())
@@ -227,22 +290,37 @@ class Transport:
         already holds the first ``offset_chunks`` chunks."""
         if meta.kind == KIND_TENSOR:
             dtype = getattr(torch, meta.dtype)
+            wire_dtype = dtype
+            numel = 1
+            for s in meta.shape:
+                numel *= s
+            casted = _should_wirecast(dtype, numel, self._wire)
+            if casted:
+                wire_dtype = self._wire
             want_cuda = meta.device_type == "cuda" and self._device is not None
             on_device = want_cuda and self._cuda_p2p
             dev = self._device if on_device else None
-            per = self._chunk_elems(torch.empty(0, dtype=dtype).element_size())
+            per = self._chunk_elems(
+                torch.empty(0, dtype=wire_dtype).element_size()
+            )
             buf = into if into is not None else torch.empty(
-                meta.shape, dtype=dtype, device=dev
+                meta.shape, dtype=wire_dtype, device=dev
             )
             if buf.numel() <= per and offset_chunks == 0:
                 chunks = [buf]
             else:
                 chunks = self._chunks(buf.view(-1), per, offset_chunks)
             ops = [dist.P2POp(dist.irecv, c, src, group=self._pg) for c in chunks]
-            if want_cuda and not on_device:
-                d = self._device
-                return ops, (lambda: buf.to(d, non_blocking=False))
-            return ops, (lambda: buf)
+
+            def finalize():
+                out = buf
+                if want_cuda and not on_device:
+                    out = out.to(self._device, non_blocking=False)
+                if casted:
+                    out = wire_unpack(out, dtype)
+                return out
+
+            return ops, finalize
         per = self._chunk_elems(1)
         buf = into if into is not None else torch.empty(meta.nbytes, dtype=torch.uint8)
         if buf.numel() <= per and offset_chunks == 0:

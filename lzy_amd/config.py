@@ -45,6 +45,7 @@ class Config:
     # channels (reference: channel-manager + slots transports)
     channel_transport: str = "rccl"    # "rccl" | "ipc"
     channel_chunk_mb: int = 256        # chunk size for large-tensor transfers
+    channel_wire_cast: str = ""        # ""|fp16|bf16|fp8e4m3|fp8e5m2: lossy wire dtype
     # HIP data-plane kernels
     hip_max_blocks: int = 0            # 0 -> kernel default grid cap
     op_streams: int = 4                # HIP streams per device for op overlap

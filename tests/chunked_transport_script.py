@@ -80,6 +80,28 @@ def main() -> None:
         assert val["k"] == 7 and val["blob"][-4:] == b"tail"
 
     dist.barrier()
+
+    # --- cast-on-the-wire: f32 payload travels as fp16 -------------------
+    Config.reset(channel_chunk_mb=1, channel_wire_cast="fp16")
+    trc = Transport(None, None, world=dist.get_world_size())
+    n2 = 1 << 18
+    payload_t = (torch.arange(n2, dtype=torch.float32) % 997) / 997.0
+    if rank == 0:
+        works, keep = trc.isend_value(payload_t, None, dst=1)
+        assert keep.dtype == torch.float16, keep.dtype  # wire form
+        for w in works:
+            w.wait()
+    else:
+        meta = describe_value("e4", torch.empty(n2, dtype=torch.float32))
+        works, fin = trc.irecv_value(meta, src=0)
+        for w in works:
+            w.wait()
+        got = fin()
+        assert got.dtype == torch.float32
+        err = (got - payload_t).abs().max().item()
+        assert err < 1e-3, f"wire-cast error {err}"
+
+    dist.barrier()
     if rank == 0:
         print("CHUNKED-TRANSPORT-OK", flush=True)
     dist.destroy_process_group()

@@ -300,3 +300,23 @@ def test_fill_pattern_mask_fused():
     assert torch.equal(a.view(torch.int16), b.view(torch.int16))
     assert torch.isfinite(a.float()).all()
     assert (a.float() >= 0).all()
+
+
+@requires_gpu
+def test_wire_pack_unpack_fp8_numerics():
+    """Cast-on-the-wire pack/unpack roundtrip via the HIP cast kernel."""
+    from lzy_amd.channels.transport import wire_pack, wire_unpack
+
+    t = torch.randn(1 << 20, device="cuda", dtype=torch.bfloat16)
+    w = wire_pack(t, torch.float8_e4m3fn)
+    assert w.dtype == torch.float8_e4m3fn and w.is_cuda
+    back = wire_unpack(w, torch.bfloat16)
+    assert back.dtype == torch.bfloat16
+    # fp8 e4m3: ~2 decimal digits; bounded relative error on randn range
+    err = (back.float() - t.float()).abs()
+    rel = (err / t.float().abs().clamp_min(1e-3)).mean().item()
+    assert rel < 0.06, f"fp8 wire roundtrip mean rel err {rel}"
+
+    w16 = wire_pack(t, torch.float16)
+    back16 = wire_unpack(w16, torch.bfloat16)
+    assert (back16.float() - t.float()).abs().max().item() < 0.01
