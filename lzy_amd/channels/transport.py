@@ -169,6 +169,25 @@ def _should_wirecast(dtype: torch.dtype, numel: int, wire) -> bool:
     )
 
 
+def _pack_contiguous(t: torch.Tensor) -> torch.Tensor:
+    """Contiguous pack for the wire.  Transposed 2-D device views go
+    through the LDS-tiled transpose kernel (coalesced both sides);
+    everything else falls back to torch's copy."""
+    if (
+        t.is_cuda
+        and t.dim() == 2
+        and t.stride(0) == 1
+        and t.stride(1) == t.shape[0]
+    ):
+        from lzy_amd import ops as _ops
+
+        if _ops.NATIVE and t.dtype in (
+            torch.float32, torch.float16, torch.bfloat16
+        ):
+            return _ops.transpose_cast(t.t())
+    return t.contiguous()
+
+
 def wire_pack(t: torch.Tensor, wire: torch.dtype) -> torch.Tensor:
     """Producer-side pack: cast to the wire dtype (HIP cast kernel on
     device, torch cast on host)."""
@@ -259,7 +278,7 @@ class Transport:
         if isinstance(value, torch.Tensor):
             t = value.detach()
             if not t.is_contiguous():
-                t = t.contiguous()
+                t = _pack_contiguous(t)
             if _should_wirecast(t.dtype, t.numel(), self._wire):
                 t = wire_pack(t, self._wire)  # cast kernel on device
             if t.is_cuda and not self._cuda_p2p:

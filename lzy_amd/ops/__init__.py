@@ -84,6 +84,11 @@ def _try_load() -> Optional[ctypes.CDLL]:
         ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p, ctypes.c_int,
         ctypes.c_int64, ctypes.c_float, ctypes.c_float, ctypes.c_void_p,
     ]
+    lib.lz_transpose_cast.restype = ctypes.c_int
+    lib.lz_transpose_cast.argtypes = [
+        ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p, ctypes.c_int,
+        ctypes.c_int64, ctypes.c_int64, ctypes.c_void_p,
+    ]
     lib.lz_axpby.restype = ctypes.c_int
     lib.lz_axpby.argtypes = [
         ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p,
@@ -326,6 +331,34 @@ def axpby(a, b, alpha: float = 1.0, beta: float = 1.0, dst=None):
             x.numel(),
             ctypes.c_float(alpha),
             ctypes.c_float(beta),
+            ctypes.c_void_p(_current_stream_ptr()),
+        )
+    )
+    return dst
+
+
+def transpose_cast(src, dst=None, dtype=None):
+    """dst = src.T (2-D), optionally casting — LDS-staged 64x64 tiles so
+    both the gather and the scatter side stay coalesced (the pack path
+    for transposed tensors; a naive strided copy runs at ~1/8 HBM rate)."""
+    import torch
+
+    lib = _require_native()
+    if not src.is_cuda or src.dim() != 2:
+        raise ValueError("transpose_cast requires a 2-D device tensor")
+    s = src.detach().contiguous()
+    rows, cols = s.shape
+    out_dtype = dtype or s.dtype
+    if dst is None:
+        dst = torch.empty(cols, rows, dtype=out_dtype, device=s.device)
+    _check(
+        lib.lz_transpose_cast(
+            ctypes.c_void_p(s.data_ptr()),
+            _dtype_code(s.dtype),
+            ctypes.c_void_p(dst.data_ptr()),
+            _dtype_code(dst.dtype),
+            rows,
+            cols,
             ctypes.c_void_p(_current_stream_ptr()),
         )
     )
