@@ -86,6 +86,8 @@ class TensorSerializer(Serializer):
         n = 1
         for s in shape:
             n *= s
+        if n == 0:  # frombuffer rejects empty buffers
+            return torch.empty(shape, dtype=dtype)
         buf = bytearray(raw)  # writable for frombuffer
         t = torch.frombuffer(buf, dtype=torch.uint8)[: n * dtype.itemsize].view(dtype)
         return t.reshape(shape).clone()

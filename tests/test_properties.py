@@ -1,0 +1,135 @@
+"""Property-based tests (hypothesis) for the invariant-heavy pieces:
+serializer round-trips, transport chunk math, journal replay, hashing."""
+import io
+
+import numpy as np
+import pytest
+import torch
+from hypothesis import given, settings, strategies as st
+
+from lzy_amd.sched import xxhash64
+from lzy_amd.serialization.registry import LzySerializerRegistry
+
+
+@st.composite
+def _json_like(draw, depth=2):
+    if depth == 0:
+        return draw(st.one_of(
+            st.integers(min_value=-(2 ** 53), max_value=2 ** 53),
+            st.floats(allow_nan=False, allow_infinity=False, width=32),
+            st.text(max_size=20),
+            st.booleans(),
+            st.none(),
+        ))
+    return draw(st.one_of(
+        st.lists(_json_like(depth=depth - 1), max_size=4),
+        st.dictionaries(st.text(max_size=8), _json_like(depth=depth - 1), max_size=4),
+        _json_like(depth=0),
+    ))
+
+
+REG = LzySerializerRegistry()
+
+
+def _roundtrip(value):
+    data, fmt = REG.dumps(value)
+    return REG.loads(data, fmt, type(value))
+
+
+@settings(max_examples=60, deadline=None)
+@given(_json_like())
+def test_serializer_roundtrip_jsonlike(value):
+    assert _roundtrip(value) == value
+
+
+@settings(max_examples=30, deadline=None)
+@given(
+    st.lists(st.floats(min_value=-1e6, max_value=1e6, allow_nan=False,
+                       width=32),
+             min_size=0, max_size=64),
+    st.sampled_from([np.float32, np.float64, np.int32]),
+)
+def test_serializer_roundtrip_numpy(xs, dtype):
+    arr = np.asarray(xs, dtype=dtype)
+    back = _roundtrip(arr)
+    assert isinstance(back, np.ndarray)
+    assert back.dtype == arr.dtype and back.shape == arr.shape
+    assert np.array_equal(back, arr)
+
+
+@settings(max_examples=30, deadline=None)
+@given(st.integers(min_value=0, max_value=10_000),
+       st.sampled_from([torch.float32, torch.int64, torch.bfloat16]))
+def test_serializer_roundtrip_tensor(n, dtype):
+    t = torch.zeros(n, dtype=dtype)
+    if n and dtype != torch.bfloat16:
+        t[0] = 7
+    back = _roundtrip(t)
+    assert back.dtype == t.dtype and back.shape == t.shape
+    assert torch.equal(back.view(torch.int8), t.view(torch.int8))
+
+
+@settings(max_examples=80, deadline=None)
+@given(st.binary(max_size=512))
+def test_xxhash64_stable_and_sensitive(data):
+    h = xxhash64(data)
+    assert 0 <= h < 2 ** 64
+    assert h == xxhash64(data)
+    if data:
+        mutated = bytes([data[0] ^ 1]) + data[1:]
+        assert xxhash64(mutated) != h
+
+
+@settings(max_examples=80, deadline=None)
+@given(
+    n=st.integers(min_value=1, max_value=1_000_000),
+    chunk_mb=st.integers(min_value=1, max_value=64),
+    elem_size=st.sampled_from([1, 2, 4, 8]),
+    offset=st.integers(min_value=0, max_value=5),
+)
+def test_transport_chunk_partition(n, chunk_mb, elem_size, offset):
+    """Chunks tile the flat buffer exactly once from any resume offset."""
+    from lzy_amd.channels.transport import Transport
+    from lzy_amd.config import Config
+
+    Config.reset(channel_chunk_mb=chunk_mb)
+    try:
+        tr = Transport(None, None, world=1)
+        per = tr._chunk_elems(elem_size)
+        flat = torch.zeros(n, dtype=torch.uint8)
+        total_chunks = (n + per - 1) // per
+        off = min(offset, total_chunks)
+        chunks = tr._chunks(flat, per, off)
+        assert len(chunks) == total_chunks - off
+        assert sum(c.numel() for c in chunks) == max(0, n - off * per)
+        for c in chunks[:-1]:
+            assert c.numel() == per
+    finally:
+        Config.reset()
+
+
+@settings(max_examples=40, deadline=None)
+@given(st.lists(
+    st.tuples(st.sampled_from(["t1", "t2", "t3"]),
+              st.sampled_from(["scheduled", "running", "done", "failed"])),
+    max_size=12,
+))
+def test_journal_replay_last_state_wins(transitions):
+    import tempfile
+
+    from lzy_amd.sched import Journal
+
+    with tempfile.TemporaryDirectory() as d:
+        path = str(__import__("pathlib").Path(d) / "x.jsonl")
+        _journal_case(path, transitions)
+
+
+def _journal_case(path, transitions):
+    from lzy_amd.sched import Journal
+    j = Journal(path)
+    want = {}
+    for tid, state in transitions:
+        j.record(tid, state)
+        want[tid] = state
+    j.close()
+    assert Journal.replay(path) == want
