@@ -152,6 +152,8 @@ class WorkerAgent:
             keeps: list = []
             for item in msg["items"]:
                 eid = item["entry"]
+                if not self.store.has(eid):  # defense: settle may be landing
+                    self.store.wait_present(eid, timeout=30.0)
                 value = self.store.get(eid)
                 ops, keep = self.transport.send_ops(
                     value, self.store.pickled.get(eid), item["dst"]
@@ -841,6 +843,9 @@ class _DriverScheduler:
                 meta.kind == KIND_BYTES
                 and meta.nbytes <= INLINE_LIMIT
                 and 0 in meta.owners
+                # owners gains rank 0 when a transfer is merely INITIATED;
+                # inline only when the value actually landed here
+                and self.pool.agent.store.has(eid)
             ):
                 specs_inline[eid] = self.pool.agent.store.pickled.get(
                     eid
@@ -855,7 +860,15 @@ class _DriverScheduler:
             )
             for r in ranks:
                 if r not in meta.owners:
-                    owner = 0 if 0 in meta.owners else next(iter(meta.owners))
+                    # source must be a CONFIRMED owner — ranks in
+                    # transferred_now were only targeted, their copy may
+                    # not have settled yet
+                    confirmed = [
+                        o for o in meta.owners
+                        if (o, eid) not in self.transferred_now
+                    ]
+                    pickable = confirmed or sorted(meta.owners)
+                    owner = 0 if 0 in pickable else pickable[0]
                     if use_ipc:
                         # zero-copy path: map the producer's HBM allocation
                         # in the consumer (hipIpc; same GPU = no copy,

@@ -1,0 +1,110 @@
+"""Randomized-DAG fuzz over the pool runtime (world N on gloo).
+
+Builds seeded random DAGs mixing scalar and tensor ops, fan-in/fan-out,
+cached ops and occasional large values, executes them on the pool
+(placement + chaining + grouped transfers + spill bookkeeping all in
+play), and checks every sink against a pure-python evaluation of the
+same DAG.  Prints FUZZ-OK on rank 0."""
+import os
+import random
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch
+
+from lzy_amd import Lzy, op
+from lzy_amd.runtime.pool import GpuPool, GpuPoolRuntime
+
+
+@op
+def src_scalar(seed: int) -> float:
+    return float((seed * 37) % 101)
+
+
+@op
+def src_tensor(seed: int, n: int) -> torch.Tensor:
+    g = torch.Generator().manual_seed(seed)
+    return torch.randint(0, 100, (n,), generator=g, dtype=torch.int64).float()
+
+
+@op
+def add(a: float, b: float) -> float:
+    return a + b
+
+
+@op
+def mul2(a: float) -> float:
+    return a * 2.0
+
+
+@op
+def tsum(t: torch.Tensor) -> float:
+    return float(t.sum().item())
+
+
+@op
+def tcombine(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    m = min(a.numel(), b.numel())
+    return a[:m] + b[:m]
+
+
+@op(cache=True, version="1.0")
+def cached_square(x: float) -> float:
+    return x * x
+
+
+def build_and_run(lzy, seed: int):
+    rng = random.Random(seed)
+    with lzy.workflow(f"fuzz-{seed}", interactive=False):
+        scalars = []  # (proxy, expected)
+        tensors = []  # (proxy, expected torch tensor on cpu)
+        for i in range(rng.randint(2, 4)):
+            s = rng.randint(0, 999)
+            scalars.append((src_scalar(s), float((s * 37) % 101)))
+        for i in range(rng.randint(1, 3)):
+            s = rng.randint(0, 999)
+            n = rng.choice([64, 1024, 100_000])
+            g = torch.Generator().manual_seed(s)
+            exp = torch.randint(0, 100, (n,), generator=g, dtype=torch.int64).float()
+            tensors.append((src_tensor(s, n), exp))
+
+        for _ in range(rng.randint(3, 10)):
+            kind = rng.random()
+            if kind < 0.35 and len(scalars) >= 2:
+                (pa, ea), (pb, eb) = rng.sample(scalars, 2)
+                scalars.append((add(pa, pb), ea + eb))
+            elif kind < 0.5:
+                p, e = rng.choice(scalars)
+                scalars.append((mul2(p), e * 2.0))
+            elif kind < 0.65:
+                p, e = rng.choice(scalars)
+                scalars.append((cached_square(p), e * e))
+            elif kind < 0.8 and len(tensors) >= 2:
+                (pa, ea), (pb, eb) = rng.sample(tensors, 2)
+                m = min(ea.numel(), eb.numel())
+                tensors.append((tcombine(pa, pb), ea[:m] + eb[:m]))
+            else:
+                p, e = rng.choice(tensors)
+                scalars.append((tsum(p), float(e.sum())))
+
+        sinks = rng.sample(scalars, min(3, len(scalars)))
+        for proxy, expected in sinks:
+            got = float(proxy)
+            assert abs(got - expected) < 1e-3 * max(1.0, abs(expected)), (
+                f"seed {seed}: got {got}, expected {expected}"
+            )
+
+
+def main() -> None:
+    GpuPool.get()
+    lzy = Lzy(runtime=GpuPoolRuntime())
+    base = int(os.environ.get("FUZZ_BASE_SEED", "1000"))
+    rounds = int(os.environ.get("FUZZ_ROUNDS", "12"))
+    for k in range(rounds):
+        build_and_run(lzy, base + k)
+    print("FUZZ-OK", flush=True)
+
+
+if __name__ == "__main__":
+    main()

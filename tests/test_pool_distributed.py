@@ -129,3 +129,16 @@ def test_chain_dispatch(tmp_path):
     r = _run_distributed("tests/pool_script_chain.py", 2, tmp_path, timeout=240)
     assert r.returncode == 0, r.stdout[-3000:] + r.stderr[-3000:]
     assert "CHAIN-OK" in r.stdout
+
+
+@pytest.mark.parametrize("world", [2, 3])
+def test_pool_fuzz_random_dags(world, tmp_path):
+    """Seeded random DAGs through the pool; every sink checked against a
+    pure-python evaluation (scheduler/chaining/transfer fuzz)."""
+    r = _run_distributed(
+        "tests/pool_script_fuzz.py", world, tmp_path,
+        extra_env={"FUZZ_BASE_SEED": str(2000 + world), "FUZZ_ROUNDS": "10"},
+        timeout=300,
+    )
+    assert r.returncode == 0, r.stdout[-3000:] + r.stderr[-3000:]
+    assert "FUZZ-OK" in r.stdout
