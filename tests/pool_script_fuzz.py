@@ -54,6 +54,18 @@ def cached_square(x: float) -> float:
     return x * x
 
 
+@op(gpu_count=2)
+def gang_sum2(x: float) -> float:
+    import torch.distributed as dist
+
+    from lzy_amd.runtime.context import op_context
+
+    ctx = op_context()
+    t = torch.tensor([x])
+    dist.all_reduce(t, group=ctx.process_group)
+    return float(t.item())
+
+
 def build_and_run(lzy, seed: int):
     rng = random.Random(seed)
     with lzy.workflow(f"fuzz-{seed}", interactive=False):
@@ -77,9 +89,12 @@ def build_and_run(lzy, seed: int):
             elif kind < 0.5:
                 p, e = rng.choice(scalars)
                 scalars.append((mul2(p), e * 2.0))
-            elif kind < 0.65:
+            elif kind < 0.6:
                 p, e = rng.choice(scalars)
                 scalars.append((cached_square(p), e * e))
+            elif kind < 0.65 and int(os.environ.get("WORLD_SIZE", "1")) >= 2:
+                p, e = rng.choice(scalars)
+                scalars.append((gang_sum2(p), e * 2.0))
             elif kind < 0.8 and len(tensors) >= 2:
                 (pa, ea), (pb, eb) = rng.sample(tensors, 2)
                 m = min(ea.numel(), eb.numel())
