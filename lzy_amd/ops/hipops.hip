@@ -359,7 +359,9 @@ __device__ __forceinline__ double lz_wave_reduce_f64(double v) {
 template <typename T, bool ABS>
 __global__ void stats_kernel(const T* __restrict__ src, int64_t n,
                              double* __restrict__ out) {
-    constexpr int V = 8;
+    // 32 B per lane per vector load regardless of dtype width: 16-bit
+    // inputs at V=8 (16 B) leave the read stream request-starved
+    constexpr int V = (sizeof(T) == 2) ? 16 : 8;
     constexpr int U = 4;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     int64_t vec_n = n / V;
