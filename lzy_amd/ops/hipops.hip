@@ -422,7 +422,7 @@ __global__ void normalize_apply_kernel(const SrcT* __restrict__ src,
     const float var = (float)(stats[1] / (double)n) - mean * mean;
     const float rstd = rsqrtf(fmaxf(var, 0.f) + eps);
 
-    constexpr int V = (sizeof(SrcT) == 2 && sizeof(DstT) == 2) ? 16 : 8;
+    constexpr int V = 8;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     int64_t vec_n = n / V;
     using SrcV = struct { SrcT v[V]; };
@@ -449,7 +449,7 @@ template <typename SrcT, typename DstT>
 __global__ void scale_shift_kernel(const SrcT* __restrict__ src,
                                    DstT* __restrict__ dst, int64_t n,
                                    float a, float b) {
-    constexpr int V = (sizeof(SrcT) == 2 && sizeof(DstT) == 2) ? 16 : 8;
+    constexpr int V = 8;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     int64_t vec_n = n / V;
     using SrcV = struct { SrcT v[V]; };
@@ -477,7 +477,7 @@ __global__ void axpby_kernel(const SrcT* __restrict__ a,
                              const SrcT* __restrict__ b,
                              DstT* __restrict__ dst, int64_t n, float alpha,
                              float beta) {
-    constexpr int V = (sizeof(SrcT) == 2 && sizeof(DstT) == 2) ? 16 : 8;
+    constexpr int V = 8;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     int64_t vec_n = n / V;
     using SrcV = struct { SrcT v[V]; };
