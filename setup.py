@@ -96,6 +96,8 @@ setup(
         "lzy_amd.channels",
         "lzy_amd.ops",
         "lzy_amd.whiteboards",
+        "lzy_amd.api",
+        "lzy_amd.api.v1",
         "lzy_amd.utils",
     ],
     ext_modules=[core_ext],
