@@ -248,3 +248,19 @@ def test_client_abort(lzy):
     except WorkflowAbortedError as e:
         assert "changed my mind" in str(e)
     assert ran == []
+
+
+def test_api_v1_compat_surface(lzy):
+    """Reference-shaped imports resolve (pylzy lzy.api.v1 parity)."""
+    from lzy_amd.api.v1 import (  # noqa: F401
+        File, GpuPoolRuntime, LocalRuntime, Lzy, LzyEnvironment,
+        LzyWorkflow, Provisioning, Runtime, cpu, env_vars, gpu,
+        lzy_auth, op as op_, whiteboard, whiteboard_,
+    )
+
+    @op_
+    def f(x: int) -> int:
+        return x * 3
+
+    with Lzy(runtime=LocalRuntime()).workflow("compat", interactive=False):
+        assert int(f(2)) == 6
