@@ -346,3 +346,46 @@ def test_configured_endpoints_served(lzy, monkeypatch):
     page = urllib.request.urlopen(f"http://127.0.0.1:{st_port}/").read().decode()
     assert "lzy-mi355x" in page
     Lzy._ENDPOINTS_SERVED = {"metrics": False, "status": False}
+
+
+@pytest.mark.parametrize("script,marker", [
+    ("examples/grid_search.py", "GRID-EXAMPLE-OK"),
+])
+def test_examples_single_process(script, marker, tmp_path):
+    import subprocess
+    import sys
+    from pathlib import Path
+
+    root = Path(__file__).resolve().parent.parent
+    env = {**os.environ, "LZY_AMD_STORAGE": str(tmp_path / "s"),
+           "PYTHONPATH": str(root)}
+    env.pop("RANK", None)
+    env.pop("WORLD_SIZE", None)
+    r = subprocess.run([sys.executable, script], cwd=root, env=env,
+                       capture_output=True, text=True, timeout=240)
+    assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
+    assert marker in r.stdout
+
+
+def test_example_ddp_two_ranks(tmp_path):
+    import socket
+    import subprocess
+    import sys
+    from pathlib import Path
+
+    root = Path(__file__).resolve().parent.parent
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    env = {**os.environ, "LZY_AMD_STORAGE": str(tmp_path / "s"),
+           "PYTHONPATH": str(root)}
+    env.pop("RANK", None)
+    env.pop("WORLD_SIZE", None)
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node=2", "--master-addr", "127.0.0.1",
+         "--master-port", str(port), "examples/train_ddp.py"],
+        cwd=root, env=env, capture_output=True, text=True, timeout=240,
+    )
+    assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
+    assert "DDP-EXAMPLE-OK" in r.stdout
