@@ -5,9 +5,10 @@ Run single-process (CPU ok):     python examples/grid_search.py
 Run one process per GPU:         python -m torch.distributed.run \
     --nnodes=1 --nproc-per-node 8 --master-addr 127.0.0.1 examples/grid_search.py
 """
+import json
 import os
 import sys
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
@@ -19,9 +20,12 @@ from lzy_amd import Lzy, op, whiteboard_
 @whiteboard_("grid_search_result")
 @dataclass
 class GridResult:
+    # whiteboard fields need STABLE serializers (same rule as the
+    # reference): containers travel as JSON text
     best_lr: float = 0.0
     best_loss: float = 0.0
-    all_scores: dict = field(default_factory=dict)
+    all_scores_json: str = "{}"
+
 
 
 @op
@@ -53,12 +57,13 @@ def main() -> None:
         best_lr = min(scores, key=scores.get)
         wb.best_lr = float(best_lr)
         wb.best_loss = scores[best_lr]
-        wb.all_scores = scores
+        wb.all_scores_json = json.dumps(scores, sort_keys=True)
         wb_id = wb.id
 
     back = lzy.whiteboard(id_=wb_id)
+    scores = json.loads(back.all_scores_json)
     print(f"best lr={back.best_lr} loss={back.best_loss:.6f} "
-          f"({len(back.all_scores)} candidates)")
+          f"({len(scores)} candidates)")
     print("GRID-EXAMPLE-OK")
 
 
