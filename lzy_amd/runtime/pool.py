@@ -781,7 +781,13 @@ class _DriverScheduler:
                     if tid not in failed_tasks:
                         failed_tasks.add(tid)
                         for ct in dag.fail(tid):
-                            self.journal.record(ct, "cancelled")
+                            if ct in self.dispatched:
+                                # chained dependent inflight on a LIVE
+                                # rank: poison its waits now
+                                failed_tasks.add(ct)
+                                self._poison_chained(ct, tid)
+                            else:
+                                self.journal.record(ct, "cancelled")
                     self.inflight -= 1
             # barrier_done etc. are routed via acks, not here
 
