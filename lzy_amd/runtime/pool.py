@@ -1028,10 +1028,15 @@ class _DriverScheduler:
             return
         rank, eids = hit
         name = self.calls[failed_tid].callable_name
-        self.pool.driver_ctrl.send(rank, {
-            "cmd": "poison", "entries": eids,
-            "reason": f"producer op {name} failed",
-        })
+        try:
+            self.pool.driver_ctrl.send(rank, {
+                "cmd": "poison", "entries": eids,
+                "reason": f"producer op {name} failed",
+            })
+        except (OSError, KeyError, BrokenPipeError):
+            # the chained child shared the dead producer's rank: its
+            # tasks were already failed by the worker_lost sweep
+            pass
 
     def _try_chain(self) -> None:
         """Dispatch tasks whose producers are all inflight/complete on one
