@@ -170,3 +170,31 @@ def test_pinned_host_spill_roundtrip():
     back_b = store.get("b")
     assert torch.equal(back_b.cpu(), ref_b)
     assert not store.spill.is_spilled("a")
+
+
+@requires_gpu
+def test_scenarios_on_gpu(tmp_path):
+    """A slice of the e2e scenario suite executed on the GPU box (ops
+    allocate device tensors through the full framework stack)."""
+    import re
+    import subprocess
+    import sys
+    from pathlib import Path
+
+    root = Path(__file__).resolve().parent.parent
+    for name in ["complex_graph", "whiteboards", "fully_cached_graph"]:
+        scen = root / "tests" / "scenarios" / name
+        env = {**__import__("os").environ,
+               "LZY_AMD_STORAGE": str(tmp_path / name),
+               "PYTHONPATH": str(root)}
+        r = subprocess.run(
+            [sys.executable, str(scen / "__init__.py")],
+            cwd=root, env=env, capture_output=True, text=True, timeout=240,
+        )
+        assert r.returncode == 0, f"{name}: {r.stdout[-1500:]}{r.stderr[-1500:]}"
+        expected = (scen / "expected_stdout").read_text().splitlines()
+        actual = [re.sub(r"^\[LZY-[^\]]*\] ", "", l)
+                  for l in r.stdout.splitlines() if l.strip()]
+        for line in expected:
+            if line != "...":
+                assert line in actual, f"{name}: missing {line!r} in {actual}"
