@@ -673,6 +673,45 @@ __global__ void transpose_cast_kernel(const SrcT* __restrict__ src,
     }
 }
 
+// v2: fully vectorized interior tiles — float4 HBM transactions on BOTH
+// sides.  Each thread loads/stores 4 consecutive elements; the LDS tile
+// keeps +1-row padding and the transposed reads gather 4 scalars from
+// 4 consecutive padded rows (65-element stride => consecutive banks).
+template <typename SrcT, typename DstT>
+__global__ void transpose_cast_kernel_v4(const SrcT* __restrict__ src,
+                                         DstT* __restrict__ dst, int64_t rows,
+                                         int64_t cols) {
+    __shared__ float tile[LZ_TT][LZ_TT + 1];
+
+    const int64_t tile_r = (int64_t)blockIdx.y * LZ_TT;
+    const int64_t tile_c = (int64_t)blockIdx.x * LZ_TT;
+
+    const int tx = threadIdx.x & 15;   // 16 threads x 4 elems = 64 cols
+    const int ty = threadIdx.x >> 4;   // 0..15 rows per pass
+    using SrcV4 = struct { SrcT v[4]; };
+    using DstV4 = struct { DstT v[4]; };
+
+    for (int r = ty; r < LZ_TT; r += 16) {
+        const int64_t gr = tile_r + r;
+        const int64_t gc = tile_c + tx * 4;
+        SrcV4 x = *reinterpret_cast<const SrcV4*>(&src[gr * cols + gc]);
+#pragma unroll
+        for (int k = 0; k < 4; ++k)
+            tile[r][tx * 4 + k] = lz_to_float<SrcT>(x.v[k]);
+    }
+    __syncthreads();
+
+    for (int c = ty; c < LZ_TT; c += 16) {
+        const int64_t gdr = tile_c + c;       // dst row = src col
+        const int64_t gdc = tile_r + tx * 4;  // dst col = src row
+        DstV4 d;
+#pragma unroll
+        for (int k = 0; k < 4; ++k)
+            d.v[k] = lz_from_float<DstT>(tile[tx * 4 + k][c]);
+        *reinterpret_cast<DstV4*>(&dst[gdr * rows + gdc]) = d;
+    }
+}
+
 extern "C" hipError_t lz_transpose_cast(const void* src, int src_dtype,
                                         void* dst, int dst_dtype,
                                         int64_t rows, int64_t cols,
@@ -680,6 +719,33 @@ extern "C" hipError_t lz_transpose_cast(const void* src, int src_dtype,
     hipStream_t s = (hipStream_t)stream;
     dim3 grid((unsigned)((cols + LZ_TT - 1) / LZ_TT),
               (unsigned)((rows + LZ_TT - 1) / LZ_TT));
+    // interior-only fast path: every tile full and 16B-aligned
+    if ((rows % LZ_TT == 0) && (cols % LZ_TT == 0)) {
+#define LZ_TCV_DST(SrcT, code, DstT)                                            \
+    case code:                                                                  \
+        hipLaunchKernelGGL((transpose_cast_kernel_v4<SrcT, DstT>), grid,        \
+                           dim3(LZ_BLOCK), 0, s, (const SrcT*)src, (DstT*)dst,  \
+                           rows, cols);                                         \
+        break;
+#define LZ_TCV_SRC(scode, SrcT)                                                 \
+    case scode:                                                                 \
+        switch (dst_dtype) {                                                    \
+            LZ_TCV_DST(SrcT, LZ_F32, float)                                     \
+            LZ_TCV_DST(SrcT, LZ_F16, __half)                                    \
+            LZ_TCV_DST(SrcT, LZ_BF16, __hip_bfloat16)                           \
+            default: return hipErrorInvalidValue;                               \
+        }                                                                       \
+        break;
+        switch (src_dtype) {
+            LZ_TCV_SRC(LZ_F32, float)
+            LZ_TCV_SRC(LZ_F16, __half)
+            LZ_TCV_SRC(LZ_BF16, __hip_bfloat16)
+            default: return hipErrorInvalidValue;
+        }
+#undef LZ_TCV_SRC
+#undef LZ_TCV_DST
+        return hipGetLastError();
+    }
 #define LZ_TC_DST(SrcT, code, DstT)                                             \
     case code:                                                                  \
         hipLaunchKernelGGL((transpose_cast_kernel<SrcT, DstT>), grid,           \
