@@ -43,6 +43,11 @@ class File:
     def copy_to(self, dst: Union[str, os.PathLike]) -> None:
         shutil.copyfile(self.path, dst)
 
+    def open(self, mode: str = "r", **kwargs):
+        """File-object access (reference tutorial:
+        ``with result.open("r") as f: ...``)."""
+        return open(self.path, mode, **kwargs)
+
     def __repr__(self) -> str:
         return f"File({self.path})"
 

@@ -264,3 +264,13 @@ def test_api_v1_compat_surface(lzy):
 
     with Lzy(runtime=LocalRuntime()).workflow("compat", interactive=False):
         assert int(f(2)) == 6
+
+
+def test_file_open_method(tmp_path):
+    from lzy_amd import File
+
+    f = File(tmp_path / "x.txt")
+    with f.open("w") as fh:
+        fh.write("via open")
+    with f.open("r") as fh:
+        assert fh.read() == "via open"
