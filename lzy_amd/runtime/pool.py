@@ -574,14 +574,39 @@ class GpuPoolRuntime(Runtime):
     def exec(self, workflow: "LzyWorkflow", calls: Sequence["LzyCall"]) -> None:
         from lzy_amd.utils.metrics import timed
 
+        # leaf tracking for finish-time persistence: results a later op
+        # never consumed may still be read AFTER the workflow exits
+        # (reference: everything lands on S3; tutorial 3 prints an op
+        # result outside the block)
+        produced = workflow.__dict__.setdefault("_produced_entries", set())
+        consumed = workflow.__dict__.setdefault("_consumed_entries", set())
+        for c in calls:
+            produced.update(c.entry_ids)
+            consumed.update(c.input_entry_ids())
+
         with timed("lzy_wf_exec"):
             sched = _DriverScheduler(self.pool, workflow, calls, self._journal)
             sched.run()
+
+    def _persist_leaves(self, workflow: "LzyWorkflow") -> None:
+        """Persist never-consumed op results to the durable tier so
+        post-exit proxy materialization finds them (fetches remote-owned
+        values through the data plane first)."""
+        produced = workflow.__dict__.get("_produced_entries", set())
+        consumed = workflow.__dict__.get("_consumed_entries", set())
+        snap = workflow.snapshot
+        for eid in sorted(produced - consumed):
+            try:
+                if not snap.storage.blob_exists(snap.get_entry(eid).storage_uri):
+                    snap.persist(eid)
+            except Exception:  # noqa: BLE001 - a failed-op leaf has no value
+                _LOG.debug("leaf %s not persisted", eid, exc_info=True)
 
     def finish(self, workflow: "LzyWorkflow") -> None:
         from lzy_amd.utils.metrics import timed
 
         with timed("lzy_wf_finish"):
+            self._persist_leaves(workflow)
             self._drop_workflow_entries(workflow)
             if self._journal is not None:
                 self._journal.close()

@@ -65,6 +65,18 @@ class PoolWb:
     label: str
 
 
+def post_exit_access() -> None:
+    """Results never touched inside the block stay readable after exit
+    (reference tutorial 3: print(data_set.DESCR) outside the workflow)."""
+    lzy = Lzy(runtime=GpuPoolRuntime())
+    with lzy.workflow("post-exit-wf"):
+        t = make_tensor(4096)
+        s = inc(41)  # also never touched inside
+    # touching AFTER the workflow: served from the durable tier
+    assert int(s) == 42, int(s)
+    assert float(t.sum().item()) == float(torch.arange(4096, dtype=torch.float32).sum())
+
+
 def main() -> None:
     # workers enter the serve loop here and never return
     GpuPool.get()
@@ -110,6 +122,8 @@ def main() -> None:
     got = lzy.whiteboard(id_=wb_id)
     assert got.total == 120.0
     assert got.label == "done"
+
+    post_exit_access()
 
     print("POOL-SCRIPT-OK", flush=True)
 
