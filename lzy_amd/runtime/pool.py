@@ -588,26 +588,18 @@ class GpuPoolRuntime(Runtime):
             sched = _DriverScheduler(self.pool, workflow, calls, self._journal)
             sched.run()
 
-    def _persist_leaves(self, workflow: "LzyWorkflow") -> None:
-        """Persist never-consumed op results to the durable tier so
-        post-exit proxy materialization finds them (fetches remote-owned
-        values through the data plane first)."""
-        produced = workflow.__dict__.get("_produced_entries", set())
-        consumed = workflow.__dict__.get("_consumed_entries", set())
-        snap = workflow.snapshot
-        for eid in sorted(produced - consumed):
-            try:
-                if not snap.storage.blob_exists(snap.get_entry(eid).storage_uri):
-                    snap.persist(eid)
-            except Exception:  # noqa: BLE001 - a failed-op leaf has no value
-                _LOG.debug("leaf %s not persisted", eid, exc_info=True)
-
     def finish(self, workflow: "LzyWorkflow") -> None:
         from lzy_amd.utils.metrics import timed
 
         with timed("lzy_wf_finish"):
-            self._persist_leaves(workflow)
-            self._drop_workflow_entries(workflow)
+            # never-consumed results may still be read after the block
+            # exits (reference tutorial 3 prints an op result outside
+            # the workflow; on S3 they simply persist) — keep the LEAF
+            # values hot and let post-exit materialization fetch them;
+            # everything consumed is dropped for HBM pressure control
+            produced = workflow.__dict__.get("_produced_entries", set())
+            consumed = workflow.__dict__.get("_consumed_entries", set())
+            self._drop_workflow_entries(workflow, keep=produced - consumed)
             if self._journal is not None:
                 self._journal.close()
                 self._journal = None
@@ -615,15 +607,19 @@ class GpuPoolRuntime(Runtime):
     def abort(self, workflow: "LzyWorkflow") -> None:
         self.finish(workflow)
 
-    def _drop_workflow_entries(self, workflow: "LzyWorkflow") -> None:
+    def _drop_workflow_entries(self, workflow: "LzyWorkflow",
+                               keep: Optional[Set[str]] = None) -> None:
         """Free this workflow's values on every rank (HBM pressure control).
 
         Persisted blobs (cache hits, whiteboard fields) survive on the
         durable tier; post-exit proxy materialization falls back to them.
-        The reference behaves the same way structurally: VM death frees
-        slot memory, S3 keeps the persisted copies.
+        ``keep`` (leaf results) stay hot for post-exit reads — the spill
+        tier bounds their HBM footprint.
         """
-        entry_ids = list(workflow.snapshot._entries.keys())
+        keep = keep or set()
+        entry_ids = [
+            e for e in workflow.snapshot._entries.keys() if e not in keep
+        ]
         if not entry_ids:
             return
         pool = self.pool
