@@ -44,6 +44,7 @@ def _default_runtime() -> Runtime:
 
     if (
         not in_op_execution()
+        and not os.environ.get("LZY_INSIDE_OP")  # op-spawned subprocess
         and os.environ.get("WORLD_SIZE")
         and int(os.environ.get("WORLD_SIZE", "1")) > 1
     ):

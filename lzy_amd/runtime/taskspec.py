@@ -199,12 +199,15 @@ def run_taskspec(
     capture = OpLogCapture.instance()
     out_buf, err_buf = capture.route_current_thread(spec.name, echo=echo_logs)
     old_env: Dict[str, Optional[str]] = {}
-    gang_env: Dict[str, str] = {}
+    # subprocesses spawned BY the op (user DDP launchers etc.) inherit
+    # this marker and must not try to become pool drivers (reference:
+    # the LZY_OP_MAIN_PID guard, startup.py:80-106)
+    gang_env: Dict[str, str] = {"LZY_INSIDE_OP": "1"}
     if spec.gang is not None:
-        gang_env = {
+        gang_env.update({
             "LZY_OP_RANK": str(spec.gang["gang_rank"]),
             "LZY_OP_WORLD_SIZE": str(len(spec.gang["ranks"])),
-        }
+        })
     try:
         for k, v in {**spec.env_vars, **gang_env}.items():
             old_env[k] = os.environ.get(k)
