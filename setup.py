@@ -100,6 +100,7 @@ setup(
         "lzy_amd.api.v1",
         "lzy_amd.utils",
     ],
+    package_data={"lzy_amd": ["py.typed"]},
     ext_modules=[core_ext],
     cmdclass={"build_ext": BuildExt},
 )
