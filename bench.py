@@ -125,11 +125,24 @@ def train_step(t: torch.Tensor) -> float:
     dtype = torch.bfloat16 if dev.type == "cuda" else torch.float32
     model, opt = _get_model(d, dev, dtype)
     x = t[: b * d].reshape(b, d).to(dtype)
-    opt.zero_grad(set_to_none=True)
-    loss = model(x).float().square().mean()
-    loss.backward()
-    opt.step()
-    return float(loss.detach().item())
+
+    def step_fn(xin: torch.Tensor) -> torch.Tensor:
+        opt.zero_grad(set_to_none=True)
+        loss = model(xin).float().square().mean()
+        loss.backward()
+        opt.step()
+        return loss.detach()
+
+    # fwd+bwd+optimizer is ~60 small launches: captured once as a
+    # hipGraph and replayed as ONE launch per step (utils/hipgraph.py)
+    global _TRAIN_GRAPH
+    try:
+        sg = _TRAIN_GRAPH
+    except NameError:
+        from lzy_amd.utils.hipgraph import StepGraph
+
+        sg = _TRAIN_GRAPH = StepGraph(step_fn)
+    return float(sg.run(x).item())
 
 
 @op

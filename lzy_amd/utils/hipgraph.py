@@ -1,0 +1,69 @@
+"""hipGraph step capture (torch.cuda.CUDAGraph IS hipGraph on ROCm).
+
+Launch-bound inner loops — a training step is ~60 small kernel launches
+— replay as ONE graph launch.  ``StepGraph`` captures a callable over
+static input buffers on first use and replays it afterwards:
+
+    sg = StepGraph(fn)             # fn(*static_inputs) -> tensor
+    out = sg.run(x, y)             # copies into static buffers, replays
+
+Capture requires static shapes/dtypes/devices; `run` re-captures if they
+change.  On CPU (or if capture fails — some op patterns are not graph
+safe) it transparently falls back to eager execution.
+"""
+from __future__ import annotations
+
+import logging
+from typing import Any, Callable, Optional, Sequence, Tuple
+
+import torch
+
+_LOG = logging.getLogger("lzy_amd.hipgraph")
+
+
+class StepGraph:
+    def __init__(self, fn: Callable[..., torch.Tensor], warmup: int = 3):
+        self._fn = fn
+        self._warmup = warmup
+        self._graph: Optional[torch.cuda.CUDAGraph] = None
+        self._static_in: Tuple[torch.Tensor, ...] = ()
+        self._static_out: Optional[torch.Tensor] = None
+        self._sig: Optional[tuple] = None
+        self.fallback_eager = not torch.cuda.is_available()
+        self.captured = False
+
+    @staticmethod
+    def _signature(args: Sequence[torch.Tensor]) -> tuple:
+        return tuple((tuple(a.shape), a.dtype, str(a.device)) for a in args)
+
+    def _capture(self, args: Sequence[torch.Tensor]) -> None:
+        self._static_in = tuple(a.clone() for a in args)
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(self._warmup):
+                out = self._fn(*self._static_in)
+        torch.cuda.current_stream().wait_stream(side)
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            self._static_out = self._fn(*self._static_in)
+        self._graph = g
+        self._sig = self._signature(args)
+        self.captured = True
+
+    def run(self, *args: torch.Tensor) -> torch.Tensor:
+        if self.fallback_eager:
+            return self._fn(*args)
+        try:
+            if self._graph is None or self._signature(args) != self._sig:
+                self._capture(args)
+            for dst, src in zip(self._static_in, args):
+                if dst.data_ptr() != src.data_ptr():
+                    dst.copy_(src, non_blocking=True)
+            self._graph.replay()
+            return self._static_out
+        except Exception as e:  # noqa: BLE001 - not all ops are graph-safe
+            _LOG.warning("hipGraph capture failed (%s); eager fallback", e)
+            self.fallback_eager = True
+            self._graph = None
+            return self._fn(*args)

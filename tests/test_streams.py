@@ -37,3 +37,23 @@ def test_workflow_on_cpu_unchanged(tmp_path, monkeypatch):
     with Lzy(runtime=LocalRuntime()).workflow("wf", interactive=False):
         r = add(torch.ones(8), torch.full((8,), 2.0))
         assert float(r.sum()) == 24.0
+
+
+def test_stepgraph_eager_fallback_cpu():
+    from lzy_amd.utils.hipgraph import StepGraph
+
+    model = torch.nn.Linear(8, 8)
+    opt = torch.optim.SGD(model.parameters(), lr=0.1)
+
+    def step(x):
+        opt.zero_grad(set_to_none=True)
+        loss = model(x).square().mean()
+        loss.backward()
+        opt.step()
+        return loss.detach()
+
+    sg = StepGraph(step)
+    a = sg.run(torch.randn(4, 8))
+    b = sg.run(torch.randn(4, 8))
+    assert a.item() >= 0 and b.item() >= 0
+    assert sg.fallback_eager and not sg.captured
