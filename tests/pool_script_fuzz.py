@@ -25,7 +25,10 @@ def src_scalar(seed: int) -> float:
 @op
 def src_tensor(seed: int, n: int) -> torch.Tensor:
     g = torch.Generator().manual_seed(seed)
-    return torch.randint(0, 100, (n,), generator=g, dtype=torch.int64).float()
+    t = torch.randint(0, 100, (n,), generator=g, dtype=torch.int64).float()
+    if torch.cuda.is_available():
+        t = t.cuda()  # device tensors exercise ipc/RCCL channels
+    return t
 
 
 @op
@@ -40,7 +43,7 @@ def mul2(a: float) -> float:
 
 @op
 def tsum(t: torch.Tensor) -> float:
-    return float(t.sum().item())
+    return float(t.float().sum().item())
 
 
 @op
