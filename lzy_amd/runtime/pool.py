@@ -184,6 +184,16 @@ class WorkerAgent:
                 {"ev": "ack", "tag": msg["tag"], "rank": self.rank,
                  "payload": data}
             )
+        elif cmd == "ipc_export_batch":
+            from lzy_amd.channels.transport import export_ipc
+
+            handles = {
+                eid: export_ipc(self.store.get(eid)) for eid in msg["entries"]
+            }
+            self.ctrl.send_event(
+                {"ev": "ack", "tag": msg["tag"], "rank": self.rank,
+                 "payload": handles}
+            )
         elif cmd == "ipc_import":
             from lzy_amd.channels.transport import import_ipc
 
@@ -855,6 +865,8 @@ class _DriverScheduler:
         wait_entries_per_rank: Dict[int, List[str]] = {r: [] for r in ranks}
         sends_by_owner: Dict[int, List[dict]] = {}
         recvs_by_rank: Dict[int, List[dict]] = {}
+        ipc_exports: Dict[int, List[str]] = {}
+        ipc_imports: List[Tuple[int, str]] = []
 
         for eid in call.input_entry_ids():
             if pending_local and eid in pending_local:
@@ -899,19 +911,12 @@ class _DriverScheduler:
                     if use_ipc:
                         # zero-copy path: map the producer's HBM allocation
                         # in the consumer (hipIpc; same GPU = no copy,
-                        # cross GPU = one xGMI DMA on import)
+                        # cross GPU = one xGMI DMA on import).  Handle
+                        # exports are batched per owner: one ack round
+                        # trip per task, not per entry.
                         if meta.ipc_handle is None:
-                            tag = f"ipc{pool.next_seq()}"
-                            pool.driver_ctrl.send(
-                                owner, {"cmd": "ipc_export", "entry": eid,
-                                        "tag": tag}
-                            )
-                            payloads = pool.wait_acks("ack", tag, [owner])
-                            meta.ipc_handle = payloads[owner]
-                        pool.driver_ctrl.send(
-                            r, {"cmd": "ipc_import", "entry": eid,
-                                "data": meta.ipc_handle}
-                        )
+                            ipc_exports.setdefault(owner, []).append(eid)
+                        ipc_imports.append((r, eid))
                         METRICS.inc("lzy_transfers_ipc")
                     else:
                         sends_by_owner.setdefault(owner, []).append(
@@ -929,6 +934,28 @@ class _DriverScheduler:
                     # transfer already initiated for an earlier task this
                     # batch; this task must still wait for its settle
                     wait_entries_per_rank[r].append(eid)
+
+        # batched hipIpc handle exchange: one export request per owner,
+        # then the imports
+        if ipc_exports:
+            tags = {}
+            for owner, eids in ipc_exports.items():
+                tag = f"ipc{pool.next_seq()}"
+                tags[owner] = (tag, eids)
+                pool.driver_ctrl.send(
+                    owner, {"cmd": "ipc_export_batch", "entries": eids,
+                            "tag": tag}
+                )
+            for owner, (tag, eids) in tags.items():
+                payloads = pool.wait_acks("ack", tag, [owner])
+                handles = payloads[owner]
+                for eid in eids:
+                    self.meta[eid].ipc_handle = handles[eid]
+        for r, eid in ipc_imports:
+            pool.driver_ctrl.send(
+                r, {"cmd": "ipc_import", "entry": eid,
+                    "data": self.meta[eid].ipc_handle}
+            )
 
         # one grouped send/recv command per rank for this task's transfers:
         # the worker issues the whole group through ONE batch_isend_irecv
