@@ -1136,6 +1136,15 @@ class _DriverScheduler:
         for wire in result.outputs:
             meta = EntryMeta.from_wire(wire)
             meta.owners = {rank}
+            inline = wire.get("inline")
+            if inline is not None and rank != 0:
+                # small result rode the completion event: land it on the
+                # driver so client materialization is a store hit, not a
+                # fetch round trip
+                self.pool.agent.store.put(
+                    meta.entry_id, unpickle_value(inline), pickled=inline
+                )
+                meta.owners.add(0)
             self.meta[meta.entry_id] = meta
             if "uri" in wire:
                 snap.update_entry_uri(meta.entry_id, wire["uri"])

@@ -138,6 +138,7 @@ class WorkerStore:
 
 
 _LOG_CAP = 256 << 10
+_INLINE_RESULT_LIMIT = 64 << 10  # small results piggyback on task_done
 
 
 def run_taskspec(
@@ -283,9 +284,14 @@ def run_taskspec(
 
         _S.record_output(eid, value, stream=stream)
         meta = describe_value(eid, value)
+        wire = meta.to_wire()
         if pickled is not None:
-            meta.nbytes = len(pickled)
-        outputs.append(meta.to_wire())
+            meta.nbytes = wire["nbytes"] = len(pickled)
+            if len(pickled) <= _INLINE_RESULT_LIMIT:
+                # small host results ride the completion event: the
+                # driver can hand them out without a fetch round trip
+                wire["inline"] = pickled
+        outputs.append(wire)
         if spec.cache and gang_primary:
             _S.wait_value(eid, value)  # D2H serialize orders after the op
             data, fmt = serializers.dumps(value)
