@@ -43,6 +43,22 @@ def gang2(x: float) -> float:
     return float(t.item())
 
 
+@op(gpu_count=2, cache=True, version="1.0")
+def gang_cached(x: float) -> float:
+    import torch.distributed as dist
+
+    from lzy_amd.runtime.context import op_context
+
+    ctx = op_context()
+    d = os.environ["LZY_AMD_STORAGE"]
+    os.makedirs(d, exist_ok=True)
+    with open(os.path.join(d, f"gangc.{os.environ.get('RANK')}"), "a") as f:
+        f.write("x")
+    t = torch.tensor([x + ctx.gang_rank])
+    dist.all_reduce(t, group=ctx.process_group)
+    return float(t.item())
+
+
 @op(gpu_count=4)
 def gang4(x: float) -> float:
     import torch.distributed as dist
@@ -78,6 +94,25 @@ def main() -> None:
         assert float(r2) == 3.0, float(r2)
         r4 = gang4(1.0)
         assert float(r4) == 10.0, float(r4)
+
+    # cached gang op: second workflow serves from the cache (no re-run)
+    import glob
+
+    marks = os.environ["LZY_AMD_STORAGE"]
+    with lzy.workflow("w4-gc1"):
+        a = gang_cached(2.0)
+        assert float(a) == 5.0, float(a)  # (2+0) + (2+1)
+    runs_before = sum(
+        len(open(f).read()) for f in glob.glob(os.path.join(marks, "gangc.*"))
+    )
+    assert runs_before >= 2, runs_before  # both gang ranks executed once
+    with lzy.workflow("w4-gc2"):
+        b = gang_cached(2.0)
+        assert float(b) == 5.0, float(b)
+    runs_after = sum(
+        len(open(f).read()) for f in glob.glob(os.path.join(marks, "gangc.*"))
+    )
+    assert runs_before == runs_after, (runs_before, runs_after)
 
     if int(os.environ.get("RANK", "0")) == 0:
         print("WORLD4-OK", flush=True)
