@@ -102,6 +102,24 @@ def main() -> None:
         assert err < 1e-3, f"wire-cast error {err}"
 
     dist.barrier()
+
+    # --- chunk-pipelined streaming pair-reduce ---------------------------
+    from lzy_amd.channels.streaming import streamed_reduce_pair
+
+    n3 = (5 << 18) + 333  # ~5 MiB f32, odd tail
+    mine = torch.arange(n3, dtype=torch.float32) + (1000.0 if rank else 0.0)
+    if rank == 0:
+        red = streamed_reduce_pair(mine, peer=1, is_receiver=True,
+                                   chunk_bytes=1 << 20)
+        want = (torch.arange(n3, dtype=torch.float32)
+                + (torch.arange(n3, dtype=torch.float32) + 1000.0)) * 0.5
+        assert torch.allclose(red, want), "streamed reduce mismatch"
+    else:
+        out = streamed_reduce_pair(mine, peer=0, is_receiver=False,
+                                   chunk_bytes=1 << 20)
+        assert out is None
+
+    dist.barrier()
     if rank == 0:
         print("CHUNKED-TRANSPORT-OK", flush=True)
     dist.destroy_process_group()
