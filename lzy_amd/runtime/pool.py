@@ -460,7 +460,8 @@ class GpuPool:
         else:
             self.events.put((rank, msg))
 
-    def wait_acks(self, kind: str, tag: str, ranks: Sequence[int], timeout: float = 300.0) -> None:
+    def wait_acks(self, kind: str, tag: str, ranks: Sequence[int],
+                  timeout: float = 300.0) -> Dict[int, Any]:
         key = f"{kind}:{tag}"
         want = set(ranks)
         with self._ack_cv:
