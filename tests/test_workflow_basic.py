@@ -274,3 +274,33 @@ def test_file_open_method(tmp_path):
         fh.write("via open")
     with f.open("r") as fh:
         assert fh.read() == "via open"
+
+
+def test_env_variable_merge_precedence(tmp_path, monkeypatch):
+    """Env merging (reference: LzyEnvironment.combine, test_op_params
+    test_workflow_env/test_op_env): Lzy < workflow < op; applied around
+    the op body and restored after."""
+    import os as _os
+
+    monkeypatch.setenv("LZY_AMD_STORAGE", str(tmp_path / "s"))
+    from lzy_amd import Lzy, LzyEnvironment
+    from lzy_amd.env.shortcuts import env_vars
+    from lzy_amd.runtime.local import LocalRuntime
+
+    seen = {}
+
+    @op(env=env_vars(A="op", C="op"))
+    def probe(x: int) -> int:
+        seen.update({k: _os.environ.get(k) for k in ("A", "B", "C", "D")})
+        return x
+
+    lzy = Lzy(runtime=LocalRuntime()).with_env_variables({"A": "lzy", "B": "lzy", "D": "lzy"})
+    wf_env = env_vars(B="wf", C="wf")
+    with lzy.workflow("envwf", env=wf_env, interactive=False):
+        int(probe(1))
+
+    assert seen["A"] == "op"    # op overrides all
+    assert seen["B"] == "wf"    # workflow overrides Lzy
+    assert seen["C"] == "op"    # op overrides workflow
+    assert seen["D"] == "lzy"   # Lzy-level reaches the op
+    assert _os.environ.get("A") != "op"  # restored afterwards
