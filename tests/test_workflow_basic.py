@@ -294,7 +294,7 @@ def test_env_variable_merge_precedence(tmp_path, monkeypatch):
         seen.update({k: _os.environ.get(k) for k in ("A", "B", "C", "D")})
         return x
 
-    lzy = Lzy(runtime=LocalRuntime()).with_env_variables({"A": "lzy", "B": "lzy", "D": "lzy"})
+    lzy = Lzy(runtime=LocalRuntime()).with_env_variables(A="lzy", B="lzy", D="lzy")
     wf_env = env_vars(B="wf", C="wf")
     with lzy.workflow("envwf", env=wf_env, interactive=False):
         int(probe(1))
