@@ -304,3 +304,21 @@ def test_env_variable_merge_precedence(tmp_path, monkeypatch):
     assert seen["C"] == "op"    # op overrides workflow
     assert seen["D"] == "lzy"   # Lzy-level reaches the op
     assert _os.environ.get("A") != "op"  # restored afterwards
+
+
+def test_container_of_proxies_as_arg(lzy):
+    """A list of lazy results passed to another op materializes correctly
+    (reference: materialize_if_sequence_of_proxies, proxy_adapter)."""
+
+    @op
+    def one(x: int) -> int:
+        return x + 1
+
+    @op
+    def total(xs: list) -> int:
+        return sum(xs)
+
+    with lzy.workflow("seq-wf", interactive=False):
+        parts = [one(i) for i in range(4)]
+        s = total(parts)
+        assert int(s) == 1 + 2 + 3 + 4
