@@ -77,6 +77,16 @@ def post_exit_access() -> None:
     assert float(t.sum().item()) == float(torch.arange(4096, dtype=torch.float32).sum())
 
 
+def container_of_proxies() -> None:
+    lzy = Lzy(runtime=GpuPoolRuntime())
+    with lzy.workflow("pool-seq-wf"):
+        parts = [inc(i) for i in range(4)]
+        t = make_tensor(8)
+        s = tensor_sum(t)
+        total = sum(int(p) for p in parts) + int(float(s))
+        assert total == (1 + 2 + 3 + 4) + 28, total
+
+
 def main() -> None:
     # workers enter the serve loop here and never return
     GpuPool.get()
@@ -124,6 +134,7 @@ def main() -> None:
     assert got.label == "done"
 
     post_exit_access()
+    container_of_proxies()
 
     print("POOL-SCRIPT-OK", flush=True)
 
