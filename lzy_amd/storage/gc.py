@@ -68,3 +68,39 @@ def gc_journals(
         except FileNotFoundError:
             continue
     return n
+
+
+def _main() -> None:  # pragma: no cover - thin CLI
+    import argparse
+    import json
+    import os
+    import tempfile
+
+    ap = argparse.ArgumentParser(description="lzy_amd storage GC")
+    ap.add_argument("--root", default=None, help="storage root (default: config)")
+    ap.add_argument("--ttl-hours", type=float, default=7 * 24.0)
+    ap.add_argument("--cache", action="store_true", help="also collect result cache")
+    ap.add_argument("--whiteboards", action="store_true")
+    args = ap.parse_args()
+    root = args.root
+    if root is None:
+        from lzy_amd.config import get_config
+
+        root = get_config().storage or os.path.join(
+            tempfile.gettempdir(), "lzy_amd_storage"
+        )
+        if root.startswith("file://"):
+            root = root[len("file://"):]
+    removed = gc_storage(
+        root, ttl_seconds=args.ttl_hours * 3600,
+        collect_cache=args.cache, collect_whiteboards=args.whiteboards,
+    )
+    removed["journals"] = gc_journals(
+        os.path.join(tempfile.gettempdir(), "lzy_amd_journal"),
+        ttl_seconds=args.ttl_hours * 3600,
+    )
+    print(json.dumps(removed))
+
+
+if __name__ == "__main__":
+    _main()
