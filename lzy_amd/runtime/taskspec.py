@@ -89,6 +89,9 @@ class WorkerStore:
             if pickled is not None:
                 self.pickled[entry_id] = pickled
             self._cond.notify_all()
+        from lzy_amd.runtime.streams import STREAMS
+
+        STREAMS.record_output(entry_id, value)  # current-stream default
         self.spill.track(entry_id, value)
         self.spill.maybe_spill(self.values)
 

@@ -141,6 +141,12 @@ class DefaultSnapshot:
     def put(self, entry_id: str, value: Any) -> None:
         self._values[entry_id] = value
         self._hashes.pop(entry_id, None)
+        # every device tensor entering the store gets a producing event
+        # (client-captured args included); executors re-record with the
+        # op's stream right after, which simply replaces this one
+        from lzy_amd.runtime.streams import STREAMS
+
+        STREAMS.record_output(entry_id, value)
         self.spill.track(entry_id, value)
         self.spill.maybe_spill(self._values)
 
