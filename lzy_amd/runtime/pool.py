@@ -299,12 +299,14 @@ class WorkerAgent:
             self._report_error(msg, e)
 
     def _report_error(self, msg: dict, e: BaseException) -> None:
+        spec = msg.get("spec")
+        where = f" in op {spec.name}" if spec is not None else f" handling {msg.get('cmd')}"
         self.ctrl.send_event(
             {
                 "ev": "agent_error",
                 "rank": self.rank,
-                "error": f"{type(e).__name__}: {e}",
-                "task_id": msg.get("spec").task_id if msg.get("spec") else None,
+                "error": f"{type(e).__name__}: {e}{where}",
+                "task_id": spec.task_id if spec is not None else None,
             }
         )
 
