@@ -214,5 +214,5 @@ def _store_outputs(call: "LzyCall", result: Any, stream=None) -> None:
         outputs = result
     for eid, value in zip(call.entry_ids, outputs):
         value = materialize(value) if is_lzy_proxy(value) else value
+        STREAMS.record_output(eid, value, stream=stream)  # pre-publication
         snap.put(eid, value)
-        STREAMS.record_output(eid, value, stream=stream)

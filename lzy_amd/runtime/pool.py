@@ -330,12 +330,13 @@ class WorkerAgent:
             pickled = None
             if not isinstance(value, torch.Tensor):
                 pickled = self.store.pickled.get(eid)
-            self.store.put(eid, value, pickled=pickled)
             # RCCL recv completion is stream-ordered, not host-ordered:
-            # publish an event so consumer op streams order after it
+            # publish an event (pre-publication) so consumer op streams
+            # order after it
             from lzy_amd.runtime.streams import STREAMS
 
             STREAMS.record_output(eid, value)
+            self.store.put(eid, value, pickled=pickled)
 
     def _run_task(self, msg: dict) -> None:
         spec: TaskSpec = msg["spec"]

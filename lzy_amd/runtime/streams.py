@@ -38,10 +38,17 @@ class StreamPlacer:
     """Round-robin HIP stream pool + entry-id -> completion-event registry."""
 
     def __init__(self) -> None:
+        from collections import deque
+
         self._lock = threading.Lock()
         self._streams: Dict[int, list] = {}  # device index -> [Stream]
         self._rr: Dict[int, int] = {}
         self._events: Dict[str, Tuple[Any, Any]] = {}  # entry -> (event, stream)
+        # entries whose event was deliberately retired after a host-side
+        # completion guarantee (workflow exit / store drop): a later read
+        # is safe and must not count as a race in strict mode
+        self._retired: set = set()
+        self._retired_order: deque = deque()
         self._tls = threading.local()
         self.enabled = torch.cuda.is_available()
 
@@ -84,15 +91,26 @@ class StreamPlacer:
     # -- producer side ------------------------------------------------------
 
     def record_output(self, entry_id: str, value: Any,
-                      stream: Optional[torch.cuda.Stream] = None) -> None:
+                      stream: Optional[torch.cuda.Stream] = None,
+                      if_absent: bool = False) -> None:
         """Record a completion event for a device-tensor entry produced on
-        ``stream`` (default: the current stream of the value's device)."""
+        ``stream`` (default: the current stream of the value's device).
+
+        ``if_absent``: keep an existing (more precise, producer-stream)
+        event — used by the blanket store-put hook, which runs on the
+        publishing thread's current stream."""
         if not self.enabled or not isinstance(value, torch.Tensor) or not value.is_cuda:
             return
+        if if_absent:
+            with self._lock:
+                if entry_id in self._events:
+                    return
         s = stream if stream is not None else torch.cuda.current_stream(value.device)
         ev = torch.cuda.Event()
         ev.record(s)
         with self._lock:
+            if if_absent and entry_id in self._events:
+                return
             self._events[entry_id] = (ev, s)
         METRICS.inc("lzy_stream_events_recorded")
 
@@ -105,8 +123,9 @@ class StreamPlacer:
             return
         with self._lock:
             hit = self._events.get(entry_id)
+            retired = entry_id in self._retired
         if hit is None:
-            if _check_mode():
+            if _check_mode() and not retired:
                 METRICS.inc("lzy_stream_races")
                 raise RuntimeError(
                     f"stream-race check: device tensor entry {entry_id} read "
@@ -119,15 +138,25 @@ class StreamPlacer:
             cur.wait_event(ev)
             METRICS.inc("lzy_stream_waits_inserted")
 
+    def _retire(self, entry_id: str) -> None:
+        # assumes self._lock held
+        if entry_id not in self._retired:
+            self._retired.add(entry_id)
+            self._retired_order.append(entry_id)
+            while len(self._retired_order) > 100_000:
+                self._retired.discard(self._retired_order.popleft())
+
     def drop(self, entry_id: str) -> None:
         with self._lock:
-            self._events.pop(entry_id, None)
+            if self._events.pop(entry_id, None) is not None:
+                self._retire(entry_id)
 
     def sync_and_drop(self, entry_id: str) -> None:
         """Host-side completion guarantee + registry cleanup (workflow
         exit: values may outlive the workflow as plain tensors)."""
         with self._lock:
             hit = self._events.pop(entry_id, None)
+            self._retire(entry_id)
         if hit is not None:
             hit[0].synchronize()
 

@@ -84,14 +84,18 @@ class WorkerStore:
         self.spill = SpillManager(device=device)
 
     def put(self, entry_id: str, value: Any, pickled: Optional[bytes] = None) -> None:
+        from lzy_amd.runtime.streams import STREAMS
+
+        # the producing event must exist BEFORE publication: a chained
+        # consumer wakes on the condition notify and immediately waits on
+        # the event.  Producers that already recorded a precise
+        # (op-stream) event win — if_absent keeps theirs.
+        STREAMS.record_output(entry_id, value, if_absent=True)
         with self._cond:
             self.values[entry_id] = value
             if pickled is not None:
                 self.pickled[entry_id] = pickled
             self._cond.notify_all()
-        from lzy_amd.runtime.streams import STREAMS
-
-        STREAMS.record_output(entry_id, value)  # current-stream default
         self.spill.track(entry_id, value)
         self.spill.maybe_spill(self.values)
 
@@ -282,10 +286,10 @@ def run_taskspec(
         pickled = None
         if not isinstance(value, torch.Tensor):
             pickled = pickle_value(value)
-        store.put(eid, value, pickled=pickled)
         from lzy_amd.runtime.streams import STREAMS as _S
 
-        _S.record_output(eid, value, stream=stream)
+        _S.record_output(eid, value, stream=stream)  # precise, pre-publication
+        store.put(eid, value, pickled=pickled)
         meta = describe_value(eid, value)
         wire = meta.to_wire()
         if pickled is not None:

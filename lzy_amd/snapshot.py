@@ -139,14 +139,14 @@ class DefaultSnapshot:
     # -- hot tier -----------------------------------------------------------
 
     def put(self, entry_id: str, value: Any) -> None:
-        self._values[entry_id] = value
-        self._hashes.pop(entry_id, None)
-        # every device tensor entering the store gets a producing event
-        # (client-captured args included); executors re-record with the
-        # op's stream right after, which simply replaces this one
+        # every device tensor entering the store has a producing event
+        # BEFORE publication (client-captured args included); precise
+        # producer-stream records made earlier win (if_absent)
         from lzy_amd.runtime.streams import STREAMS
 
-        STREAMS.record_output(entry_id, value)
+        STREAMS.record_output(entry_id, value, if_absent=True)
+        self._values[entry_id] = value
+        self._hashes.pop(entry_id, None)
         self.spill.track(entry_id, value)
         self.spill.maybe_spill(self._values)
 

@@ -188,13 +188,13 @@ class SpillManager:
             return None
         host, device, _ = hit
         t = self._mover.to_device(host, device)
-        store_values[entry_id] = t
-        self.track(entry_id, t)
-        # re-publish a producing event (the H2D is ordered into the
-        # current stream by the mover; later consumers wait on this)
+        # re-publish a producing event BEFORE publication (the H2D is
+        # ordered into the current stream by the mover)
         from lzy_amd.runtime.streams import STREAMS
 
         if isinstance(t, torch.Tensor) and t.is_cuda:
             STREAMS.record_output(entry_id, t)
+        store_values[entry_id] = t
+        self.track(entry_id, t)
         METRICS.inc("lzy_spill_in")
         return t
