@@ -13,7 +13,7 @@ Topology:
     and is ALSO a worker (its GPU is not idle).
   * ranks 1..N-1 = workers: enter the serve loop at pool init and execute
     tasks / transfers until shutdown (then sys.exit(0)).
-  * control plane: one TCP connection per worker to the driver (star).
+  * control plane: one AF_UNIX connection per worker to the driver (star).
   * data plane: torch.distributed isend/irecv on a dedicated process
     group — RCCL over xGMI for device tensors, gloo for CPU bytes;
     transfer commands carry a driver-assigned global sequence so every
