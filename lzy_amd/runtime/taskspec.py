@@ -302,8 +302,10 @@ def run_taskspec(
         if spec.cache and gang_primary:
             _S.wait_value(eid, value)  # D2H serialize orders after the op
             data, fmt = serializers.dumps(value)
-            storage.write_bytes(out_uris[eid], data)
+            # sidecar BEFORE the data blob: cache hits gate on the data
+            # blob's existence, so the format must already be readable
             _write_fmt(storage, out_uris[eid], fmt)
+            storage.write_bytes(out_uris[eid], data)
 
     METRICS.inc("lzy_op_runs", op=spec.name)
     elapsed = time.perf_counter() - t0

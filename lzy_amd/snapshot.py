@@ -219,8 +219,9 @@ class DefaultSnapshot:
             entry.storage_uri
         ):
             return entry.storage_uri
-        self._storage.write_bytes(entry.storage_uri, data)
+        # sidecar first: concurrent readers gate on the data blob
         self._storage.write_bytes(entry.storage_uri + ".fmt", fmt.encode())
+        self._storage.write_bytes(entry.storage_uri, data)
         self._hashes[entry_id] = h
         return entry.storage_uri
 
