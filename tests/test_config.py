@@ -71,3 +71,27 @@ def test_reset_overrides_pin_until_next_reset(monkeypatch):
     assert Config.get().channel_chunk_mb == 48  # explicit override pinned
     Config.reset()
     assert Config.get().channel_chunk_mb == 96
+
+
+def test_bad_yaml_rejected(tmp_path, monkeypatch):
+    p = tmp_path / "bad.yaml"
+    p.write_text("- just\n- a list\n")
+    monkeypatch.setenv("LZY_CONFIG", str(p))
+    with pytest.raises(ValueError, match="mapping"):
+        Config.resolve()
+
+
+def test_unknown_yaml_keys_ignored(tmp_path, monkeypatch):
+    p = tmp_path / "extra.yaml"
+    p.write_text("storage: /x\nnot_a_real_knob: 1\n")
+    monkeypatch.setenv("LZY_CONFIG", str(p))
+    cfg = Config.resolve()
+    assert cfg.storage == "/x"
+    assert not hasattr(cfg, "not_a_real_knob")
+
+
+def test_bool_coercions(monkeypatch):
+    for raw, want in [("1", True), ("true", True), ("on", True),
+                      ("0", False), ("no", False), ("off", False)]:
+        monkeypatch.setenv("LZY_CACHE_ENABLED", raw)
+        assert Config.resolve().cache_enabled is want
