@@ -13,6 +13,7 @@ def _isolate_config(monkeypatch):
         if k.startswith("LZY_"):
             monkeypatch.delenv(k, raising=False)
     yield
+    os.environ.pop("LZY_CONFIG", None)
     Config.reset()
 
 
