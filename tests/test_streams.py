@@ -57,3 +57,27 @@ def test_stepgraph_eager_fallback_cpu():
     b = sg.run(torch.randn(4, 8))
     assert a.item() >= 0 and b.item() >= 0
     assert sg.fallback_eager and not sg.captured
+
+
+def test_ops_cpu_guards():
+    """Device-only ops fail loudly with typed errors on host tensors —
+    no silent eager fallback (driver round-end checks this posture)."""
+    import pytest as _pytest
+
+    import lzy_amd.ops as ops
+
+    t = torch.ones(1024)
+    if not ops.NATIVE:
+        _pytest.skip("native lib not built")
+    with _pytest.raises(ValueError):
+        ops.device_checksum(t)
+    with _pytest.raises(ValueError):
+        ops.normalize(t)
+    with _pytest.raises(ValueError):
+        ops.scale_shift(t, 1.0, 0.0)
+    with _pytest.raises(ValueError):
+        ops.axpby(t, t)
+    with _pytest.raises(ValueError):
+        ops.transpose_cast(t.view(32, 32))
+    with _pytest.raises(ValueError):
+        ops.stats(t)
