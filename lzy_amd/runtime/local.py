@@ -60,11 +60,19 @@ class LocalRuntime(Runtime):
         self._journal_dir = journal_dir
         self._pool: Optional[ThreadPoolExecutor] = None
         self._journal: Optional[Journal] = None
+        # single-flight: one workflow at a time per runtime instance —
+        # concurrent client threads serialize here instead of racing the
+        # journal/pool state (the reference serializes per execution via
+        # its DB; nested workflows use their own Lzy()/runtime)
+        import threading as _threading
+
+        self._flight = _threading.Lock()
 
     def storage(self) -> Optional[StorageConfig]:
         return StorageConfig(uri=default_storage_uri())
 
     def start(self, workflow: "LzyWorkflow") -> None:
+        self._flight.acquire()
         self._pool = ThreadPoolExecutor(
             max_workers=self._max_workers, thread_name_prefix="lzy-op"
         )
@@ -168,4 +176,9 @@ class LocalRuntime(Runtime):
         if self._journal is not None:
             self._journal.close()
             self._journal = None
+        if self._flight.locked():
+            try:
+                self._flight.release()
+            except RuntimeError:
+                pass
         OpLogCapture.instance().uninstall()

@@ -541,6 +541,8 @@ class GpuPoolRuntime(Runtime):
         )
         self._journal: Optional[Journal] = None
         self._pool: Optional[GpuPool] = None
+        # single-flight: the driver scheduler state is per-workflow
+        self._flight = threading.Lock()
 
     @property
     def pool(self) -> GpuPool:
@@ -554,6 +556,7 @@ class GpuPoolRuntime(Runtime):
     def start(self, workflow: "LzyWorkflow") -> None:
         from lzy_amd.utils.metrics import timed
 
+        self._flight.acquire()  # one workflow at a time per runtime
         with timed("lzy_wf_start"):
             self._start(workflow)
 
@@ -603,6 +606,18 @@ class GpuPoolRuntime(Runtime):
             sched.run()
 
     def finish(self, workflow: "LzyWorkflow") -> None:
+        from lzy_amd.utils.metrics import timed
+
+        try:
+            self._finish_inner(workflow)
+        finally:
+            if self._flight.locked():
+                try:
+                    self._flight.release()
+                except RuntimeError:
+                    pass
+
+    def _finish_inner(self, workflow: "LzyWorkflow") -> None:
         from lzy_amd.utils.metrics import timed
 
         with timed("lzy_wf_finish"):

@@ -322,3 +322,31 @@ def test_container_of_proxies_as_arg(lzy):
         parts = [one(i) for i in range(4)]
         s = total(parts)
         assert int(s) == 1 + 2 + 3 + 4
+
+
+def test_concurrent_client_threads_serialize(tmp_path, monkeypatch):
+    """Two client threads sharing one runtime run their workflows
+    one-at-a-time (single-flight) with correct results."""
+    import threading
+
+    monkeypatch.setenv("LZY_AMD_STORAGE", str(tmp_path / "s"))
+    from lzy_amd import Lzy
+    from lzy_amd.runtime.local import LocalRuntime
+
+    lzy = Lzy(runtime=LocalRuntime())
+    results = {}
+
+    @op
+    def mul(a: int, b: int) -> int:
+        return a * b
+
+    def client(k: int):
+        with lzy.workflow(f"thr-{k}", interactive=False):
+            results[k] = int(mul(k, 10))
+
+    ths = [threading.Thread(target=client, args=(k,)) for k in range(4)]
+    for t in ths:
+        t.start()
+    for t in ths:
+        t.join(timeout=60)
+    assert results == {0: 0, 1: 10, 2: 20, 3: 30}
