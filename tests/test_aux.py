@@ -389,3 +389,40 @@ def test_example_ddp_two_ranks(tmp_path):
     )
     assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
     assert "DDP-EXAMPLE-OK" in r.stdout
+
+
+def test_doctor_cli(tmp_path):
+    import json as _json
+    import subprocess
+    import sys
+    from pathlib import Path
+
+    root = Path(__file__).resolve().parent.parent
+    r = subprocess.run(
+        [sys.executable, "-m", "lzy_amd"],
+        cwd=root, env={**os.environ, "PYTHONPATH": str(root)},
+        capture_output=True, text=True, timeout=120,
+    )
+    assert r.returncode == 0, r.stderr[-1000:]
+    rep = _json.loads(r.stdout)
+    assert rep["native"]["sched_core_cpp"] is True
+    assert "channel_transport" in rep["config"]
+
+
+def test_failed_state_in_executions(lzy):
+    from lzy_amd.exceptions import LzyExecutionError
+
+    @op
+    def nope(x: int) -> int:
+        raise RuntimeError("fail here")
+
+    try:
+        with lzy.workflow("failing-exec-wf"):
+            int(nope(1))
+    except LzyExecutionError:
+        pass
+    e = next(
+        x for x in lzy.executions()
+        if x["execution_id"].startswith("failing-exec-wf")
+    )
+    assert "failed" in e["states"]
