@@ -87,6 +87,19 @@ def container_of_proxies() -> None:
         assert total == (1 + 2 + 3 + 4) + 28, total
 
 
+def client_abort() -> None:
+    from lzy_amd.exceptions import WorkflowAbortedError
+
+    lzy = Lzy(runtime=GpuPoolRuntime())
+    try:
+        with lzy.workflow("pool-abort-wf") as wf:
+            inc(1)  # queued, dropped by abort
+            wf.abort("operator said stop")
+        raise AssertionError("expected WorkflowAbortedError")
+    except WorkflowAbortedError as e:
+        assert "operator said stop" in str(e)
+
+
 def main() -> None:
     # workers enter the serve loop here and never return
     GpuPool.get()
@@ -135,6 +148,7 @@ def main() -> None:
 
     post_exit_access()
     container_of_proxies()
+    client_abort()
 
     print("POOL-SCRIPT-OK", flush=True)
 
