@@ -98,6 +98,9 @@ def client_abort() -> None:
         raise AssertionError("expected WorkflowAbortedError")
     except WorkflowAbortedError as e:
         assert "operator said stop" in str(e)
+    # the pool survives an abort: next workflow runs normally
+    with lzy.workflow("pool-after-abort"):
+        assert int(inc(10)) == 11
 
 
 def main() -> None:
