@@ -57,7 +57,8 @@ def main() -> None:
     t0 = time.perf_counter()
     try:
         with lzy.workflow("chain-fail-wf"):
-            y = consumer(slow_boom(1))
+            # A(fails) -> B -> C: the whole chained suffix is poisoned
+            y = consumer(consumer(slow_boom(1)))
             int(y)
         raise AssertionError("expected failure")
     except LzyExecutionError as e:
