@@ -84,11 +84,12 @@ def main() -> None:
     # --- cast-on-the-wire: f32 payload travels as fp16 -------------------
     Config.reset(channel_chunk_mb=1, channel_wire_cast="fp16")
     trc = Transport(None, None, world=dist.get_world_size())
-    n2 = 1 << 18
+    n2 = (1 << 20) + 77  # fp16 wire ~2 MiB -> multiple 1 MiB chunks + tail
     payload_t = (torch.arange(n2, dtype=torch.float32) % 997) / 997.0
     if rank == 0:
         works, keep = trc.isend_value(payload_t, None, dst=1)
         assert keep.dtype == torch.float16, keep.dtype  # wire form
+        assert len(works) >= 3, len(works)  # chunked on the wire
         for w in works:
             w.wait()
     else:
