@@ -24,7 +24,3 @@ def lzy(storage_root):
 
     return Lzy(runtime=LocalRuntime())
 
-
-def pytest_configure(config):
-    # belt-and-braces with pytest.ini: the driver filters on -m gpu
-    config.addinivalue_line("markers", "gpu: test requires an MI355X GPU")
