@@ -133,3 +133,35 @@ def _journal_case(path, transitions):
         want[tid] = state
     j.close()
     assert Journal.replay(path) == want
+
+
+@settings(max_examples=60, deadline=None)
+@given(
+    dtype=st.sampled_from([torch.float32, torch.bfloat16, torch.float16,
+                           torch.int64, torch.uint8]),
+    numel=st.integers(min_value=0, max_value=1 << 22),
+    wire=st.sampled_from(["", "fp16", "bf16", "fp8e4m3", "fp8e5m2"]),
+)
+def test_wirecast_decision_deterministic_and_sound(dtype, numel, wire):
+    """Sender and receiver derive the SAME wire decision from
+    (dtype, numel, config) — and never 'cast' to a wider/equal dtype."""
+    from lzy_amd.channels.transport import _WIRE_DTYPES, _should_wirecast
+
+    w = _WIRE_DTYPES.get(wire)
+    a = _should_wirecast(dtype, numel, w)
+    b = _should_wirecast(dtype, numel, w)
+    assert a == b
+    if a:
+        assert w is not None
+        assert torch.empty(0, dtype=w).element_size() < torch.empty(
+            0, dtype=dtype
+        ).element_size()
+        assert numel >= (1 << 16)
+        assert dtype.is_floating_point
+
+
+@settings(max_examples=40, deadline=None)
+@given(st.binary(min_size=0, max_size=4096))
+def test_serializer_bytes_roundtrip(data):
+    back = _roundtrip(data)
+    assert back == data
