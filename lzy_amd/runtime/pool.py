@@ -408,12 +408,17 @@ class GpuPool:
             if not dist.is_initialized():
                 # explicit composite: CPU collectives on gloo (control
                 # broadcasts, barriers), CUDA tensors on nccl (=RCCL)
+                import datetime as _dt
+
                 backend = "cpu:gloo,cuda:nccl" if torch.cuda.is_available() else "gloo"
                 dist.init_process_group(
                     backend=backend,
                     rank=self.rank,
                     world_size=self.world,
                     device_id=self.device if cuda_ok else None,
+                    # fail fast instead of the 30-min default: a broken
+                    # rendezvous should error, not hang the harness
+                    timeout=_dt.timedelta(seconds=180),
                 )
             pg_data = dist.new_group()  # dedicated transfer group
 
