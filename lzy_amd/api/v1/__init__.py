@@ -5,7 +5,7 @@ from lzy_amd.core.lzy import Lzy, lzy_auth
 from lzy_amd.core.op import op
 from lzy_amd.core.workflow import LzyWorkflow
 from lzy_amd.env.environment import LzyEnvironment
-from lzy_amd.env.provisioning import Provisioning
+from lzy_amd.env.provisioning import GpuType, Provisioning
 from lzy_amd.env.shortcuts import (
     auto_python,
     cpu,
@@ -23,7 +23,7 @@ from lzy_amd.whiteboards.wb import whiteboard, whiteboard_
 
 __all__ = [
     "Lzy", "lzy_auth", "op", "LzyWorkflow", "LzyEnvironment",
-    "Provisioning", "Runtime", "LocalRuntime", "GpuPoolRuntime", "File",
+    "Provisioning", "GpuType", "Runtime", "LocalRuntime", "GpuPoolRuntime", "File",
     "whiteboard", "whiteboard_", "gpu", "cpu", "ram", "env_vars",
     "auto_python", "manual_python", "docker_image",
 ]
