@@ -63,8 +63,11 @@ class Provisioning:
             return 1
         return 0
 
-    def resolve_pool(self, pools: Sequence[PoolSpec]) -> PoolSpec:
-        """Pick the cheapest feasible pool (reference resolve_pool analogue)."""
+    def resolve_pool(self, pools: Sequence[PoolSpec], score=None) -> PoolSpec:
+        """Pick the best feasible pool (reference: score-based resolve,
+        provisioning.py:126).  ``score(provisioning, pool) -> float`` may
+        override the default cheapest-fit ranking (higher wins), matching
+        the reference's pluggable ``score`` function."""
         need_gpus = self.effective_gpu_count
         feasible = [
             p
@@ -77,6 +80,8 @@ class Provisioning:
             raise BadProvisioningError(
                 f"No pool satisfies provisioning {self} (available: {list(pools)})"
             )
+        if score is not None:
+            return max(feasible, key=lambda p: score(self, p))
         # cheapest = fewest GPUs then fewest CPUs
         return min(feasible, key=lambda p: (p.gpu_count, p.cpu_count))
 

@@ -426,3 +426,17 @@ def test_failed_state_in_executions(lzy):
         if x["execution_id"].startswith("failing-exec-wf")
     )
     assert "failed" in e["states"]
+
+
+def test_provisioning_custom_score():
+    """Pluggable score function (reference: lzy.env.provisioning.score)."""
+    from lzy_amd.env.provisioning import Provisioning, node_pools
+
+    pools = node_pools(8)
+    # prefer the LARGEST feasible pool instead of the cheapest
+    big = Provisioning(gpu_count=1).resolve_pool(
+        pools, score=lambda prov, p: p.gpu_count
+    )
+    assert big.gpu_count == 8
+    cheap = Provisioning(gpu_count=1).resolve_pool(pools)
+    assert cheap.gpu_count == 1
