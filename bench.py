@@ -286,6 +286,14 @@ def main() -> None:
             "shard_mb": shard_mb,
             "ops_per_dag": n_ops,
             "dispatch_p99_ms": round(1000.0 * dispatch.get("p99", 0.0), 4),
+            "transfers": int(METRICS.counter_value("lzy_transfers")
+                             + METRICS.counter_value("lzy_transfers_ipc")),
+            "transfer_gb": round(
+                METRICS.counter_value("lzy_transfer_bytes") / 1e9, 3
+            ),
+            "chain_dispatches": int(
+                METRICS.counter_value("lzy_chain_dispatches")
+            ),
         },
     }
     print(json.dumps(result), flush=True)
