@@ -170,6 +170,7 @@ class LocalRuntime(Runtime):
         self._teardown()
 
     def _teardown(self) -> None:
+        torn = self._pool is not None
         if self._pool is not None:
             self._pool.shutdown(wait=True)
             self._pool = None
@@ -181,4 +182,5 @@ class LocalRuntime(Runtime):
                 self._flight.release()
             except RuntimeError:
                 pass
-        OpLogCapture.instance().uninstall()
+        if torn:  # balance exactly one uninstall per start
+            OpLogCapture.instance().uninstall()

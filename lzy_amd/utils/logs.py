@@ -85,6 +85,8 @@ class OpLogCapture:
 
     def uninstall(self) -> None:
         with self._lock:
+            if self._depth == 0:
+                return  # already uninstalled (double teardown is benign)
             self._depth -= 1
             if self._depth == 0 and self._out_tee is not None:
                 sys.stdout = self._out_tee._fallback  # type: ignore[assignment]
