@@ -86,8 +86,16 @@ class LzyWorkflow:
     def __exit__(self, exc_type, exc_val, exc_tb) -> bool:
         try:
             if exc_type is None:
-                self.barrier()
-                self._finalize_whiteboards()
+                try:
+                    self.barrier()
+                    self._finalize_whiteboards()
+                except BaseException:
+                    # exit-time failures (op error first surfacing at the
+                    # final barrier, unassigned whiteboard field) must
+                    # still tear the runtime down — otherwise the
+                    # single-flight lock and log capture leak
+                    self.owner.runtime.abort(self)
+                    raise
                 self.owner.runtime.finish(self)
             else:
                 _LOG.warning("Workflow %s aborted: %s", self.name, exc_val)

@@ -350,3 +350,29 @@ def test_concurrent_client_threads_serialize(tmp_path, monkeypatch):
     for t in ths:
         t.join(timeout=60)
     assert results == {0: 0, 1: 10, 2: 20, 3: 30}
+
+
+def test_runtime_survives_exit_time_failures(lzy):
+    """Failures surfacing at the exit barrier (or whiteboard finalize)
+    still release the runtime: the next workflow runs normally."""
+    from lzy_amd.exceptions import LzyExecutionError
+
+    @op
+    def boom(x: int) -> int:
+        raise RuntimeError("late boom")
+
+    @op
+    def fine(x: int) -> int:
+        return x + 1
+
+    # op failure that is never materialized in the body -> raises at exit
+    try:
+        with lzy.workflow("late-fail-wf", interactive=False):
+            boom(1)  # lazy, untouched
+        raise AssertionError("expected LzyExecutionError")
+    except LzyExecutionError:
+        pass
+
+    # the SAME runtime must accept the next workflow (no leaked lock)
+    with lzy.workflow("after-late-fail", interactive=False):
+        assert int(fine(1)) == 2
