@@ -3,10 +3,11 @@
 Two jobs:
 
 1. **Overlap** (BASELINE north star: "overlaps independent ops on separate
-   HIP streams"): each op executing in the in-process runtime is placed on
-   one of a small pool of HIP streams per device, round-robin, so
-   independent ops on the *same* GPU overlap their kernels/DMA instead of
-   serializing on the legacy default stream.
+   HIP streams"): each executor THREAD is bound to one of a small pool of
+   HIP streams per device (round-robin at first use), so concurrently
+   executing ops on the *same* GPU overlap their kernels/DMA instead of
+   serializing on the legacy default stream — see next_stream() for why
+   per-thread beats per-op rotation.
 
 2. **Race detection** (SURVEY §5.2 — the reference is JVM and needs none;
    a HIP runtime does): cross-stream value hand-off is only safe through
