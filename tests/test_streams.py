@@ -81,3 +81,19 @@ def test_ops_cpu_guards():
         ops.transpose_cast(t.view(32, 32))
     with _pytest.raises(ValueError):
         ops.stats(t)
+
+
+def test_record_if_absent_keeps_precise_event():
+    """if_absent must not clobber a precise producer record (the put-hook
+    contract behind the publication-ordering fix)."""
+    import torch as _t
+
+    from lzy_amd.runtime.streams import StreamPlacer
+
+    p = StreamPlacer()
+    if not p.enabled:
+        # CPU: both calls no-op; just exercise the API
+        p.record_output("e", _t.ones(2), if_absent=True)
+        p.record_output("e", _t.ones(2))
+        assert "e" not in p._events
+        return
