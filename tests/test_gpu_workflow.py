@@ -227,3 +227,18 @@ def test_stepgraph_captures_on_gpu():
     for _ in range(20):
         last = float(sg.run(xs).item())
     assert last < first
+
+
+@requires_gpu
+def test_record_if_absent_gpu_contract():
+    from lzy_amd.runtime.streams import StreamPlacer
+
+    p = StreamPlacer()
+    t = torch.ones(8, device="cuda")
+    s1 = torch.cuda.Stream()
+    p.record_output("e", t, stream=s1)          # precise producer record
+    ev1 = p._events["e"][0]
+    p.record_output("e", t, if_absent=True)     # blanket hook: must keep
+    assert p._events["e"][0] is ev1
+    p.record_output("e", t)                     # explicit: may replace
+    assert p._events["e"][0] is not ev1
