@@ -150,6 +150,9 @@ _WIRE_DTYPES = {
 }
 _WIRE_CASTABLE = (torch.float32, torch.bfloat16, torch.float16)
 _WIRE_MIN_ELEMS = 1 << 16
+# wire dtypes the gloo path can carry: float8 isend/irecv raises on gloo,
+# so fp8 wire-cast is RCCL-device-path only
+_GLOO_WIRE_OK = (torch.float16, torch.bfloat16)
 
 
 def wire_cast_dtype() -> Optional[torch.dtype]:
@@ -239,6 +242,16 @@ class Transport:
         self._chunk_bytes = get_config().channel_chunk_mb << 20
         self._wire = wire_cast_dtype()
 
+    def _wire_allowed(self, producer_cuda: bool) -> bool:
+        """May the configured wire dtype be used for this transfer?  The
+        RCCL device path carries anything; the gloo path (CPU values,
+        host-staged device tensors) only fp16/bf16."""
+        if self._wire is None:
+            return False
+        if producer_cuda and self._cuda_p2p:
+            return True
+        return self._wire in _GLOO_WIRE_OK
+
     # -- chunking ------------------------------------------------------------
     #
     # Large values go over the wire as a sequence of fixed-size chunks
@@ -279,7 +292,13 @@ class Transport:
             t = value.detach()
             if not t.is_contiguous():
                 t = _pack_contiguous(t)
-            if _should_wirecast(t.dtype, t.numel(), self._wire):
+            # wire-cast decision must be symmetric with recv_ops: both
+            # sides derive it from (producer device, _cuda_p2p, config).
+            # fp8 wire dtypes are RCCL-device-path only — gloo (CPU
+            # tensors, host-staged fallback) cannot carry float8.
+            if self._wire_allowed(t.is_cuda) and _should_wirecast(
+                t.dtype, t.numel(), self._wire
+            ):
                 t = wire_pack(t, self._wire)  # cast kernel on device
             if t.is_cuda and not self._cuda_p2p:
                 t = t.cpu()
@@ -313,11 +332,14 @@ class Transport:
             numel = 1
             for s in meta.shape:
                 numel *= s
-            casted = _should_wirecast(dtype, numel, self._wire)
-            if casted:
-                wire_dtype = self._wire
             want_cuda = meta.device_type == "cuda" and self._device is not None
             on_device = want_cuda and self._cuda_p2p
+            # mirror of the sender's decision (same predicate, same inputs)
+            casted = self._wire_allowed(
+                meta.device_type == "cuda"
+            ) and _should_wirecast(dtype, numel, self._wire)
+            if casted:
+                wire_dtype = self._wire
             dev = self._device if on_device else None
             per = self._chunk_elems(
                 torch.empty(0, dtype=wire_dtype).element_size()

@@ -55,6 +55,9 @@ class Config:
     chain_dispatch: bool = True        # eager same-rank dependency dispatch
     gang_timeout_s: float = 120.0      # gang-allocation wait bound
     heartbeat_period_s: float = 2.0    # worker liveness probe period
+    task_retries: int = 1              # re-dispatches per task after worker death
+    settle_wait_s: float = 120.0       # worker wait for an inbound entry to land
+    ack_wait_s: float = 300.0          # driver wait for worker acks
     # result cache / snapshot
     cache_enabled: bool = True
     # HBM store spill tier (pinned-host async)

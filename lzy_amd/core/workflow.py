@@ -165,7 +165,9 @@ class LzyWorkflow:
             f"Workflow '{self.name}': execute graph of {len(names)} op(s) "
             f"[{preview}]? (y/n) "
         ).strip().lower()
-        if answer not in ("y", "yes", ""):
+        # default-deny: a bare Enter declines (reference interactive
+        # confirm defaults to 'No', remote runtime.py:424-434)
+        if answer not in ("y", "yes"):
             raise WorkflowAbortedError(
                 f"workflow {self.name} execution declined by user"
             )

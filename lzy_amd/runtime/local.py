@@ -67,12 +67,17 @@ class LocalRuntime(Runtime):
         import threading as _threading
 
         self._flight = _threading.Lock()
+        # Lock has no ownership: remember WHICH workflow holds the flight
+        # so finish/abort of a never-started (or already-finished) one
+        # cannot release another thread's active flight.
+        self._flight_owner: Optional[str] = None
 
     def storage(self) -> Optional[StorageConfig]:
         return StorageConfig(uri=default_storage_uri())
 
     def start(self, workflow: "LzyWorkflow") -> None:
         self._flight.acquire()
+        self._flight_owner = workflow.execution_id
         self._pool = ThreadPoolExecutor(
             max_workers=self._max_workers, thread_name_prefix="lzy-op"
         )
@@ -164,12 +169,14 @@ class LocalRuntime(Runtime):
                 )
 
     def finish(self, workflow: "LzyWorkflow") -> None:
-        self._teardown()
+        self._teardown(workflow)
 
     def abort(self, workflow: "LzyWorkflow") -> None:
-        self._teardown()
+        self._teardown(workflow)
 
-    def _teardown(self) -> None:
+    def _teardown(self, workflow: "LzyWorkflow") -> None:
+        if self._flight_owner != workflow.execution_id:
+            return  # this workflow never started (or already finished)
         torn = self._pool is not None
         if self._pool is not None:
             self._pool.shutdown(wait=True)
@@ -177,10 +184,10 @@ class LocalRuntime(Runtime):
         if self._journal is not None:
             self._journal.close()
             self._journal = None
-        if self._flight.locked():
-            try:
-                self._flight.release()
-            except RuntimeError:
-                pass
+        self._flight_owner = None
+        try:
+            self._flight.release()
+        except RuntimeError:
+            pass
         if torn:  # balance exactly one uninstall per start
             OpLogCapture.instance().uninstall()

@@ -546,8 +546,13 @@ class GpuPoolRuntime(Runtime):
         )
         self._journal: Optional[Journal] = None
         self._pool: Optional[GpuPool] = None
-        # single-flight: the driver scheduler state is per-workflow
+        # single-flight: the driver scheduler state is per-workflow.
+        # threading.Lock has no ownership, so finish/abort track WHICH
+        # workflow holds the flight and only release for that one — a
+        # stray abort on a never-started workflow must not unlock a
+        # different thread's active flight.
         self._flight = threading.Lock()
+        self._flight_owner: Optional[str] = None
 
     @property
     def pool(self) -> GpuPool:
@@ -562,6 +567,7 @@ class GpuPoolRuntime(Runtime):
         from lzy_amd.utils.metrics import timed
 
         self._flight.acquire()  # one workflow at a time per runtime
+        self._flight_owner = workflow.execution_id
         with timed("lzy_wf_start"):
             self._start(workflow)
 
@@ -616,7 +622,8 @@ class GpuPoolRuntime(Runtime):
         try:
             self._finish_inner(workflow)
         finally:
-            if self._flight.locked():
+            if self._flight_owner == workflow.execution_id:
+                self._flight_owner = None
                 try:
                     self._flight.release()
                 except RuntimeError:
@@ -625,6 +632,8 @@ class GpuPoolRuntime(Runtime):
     def _finish_inner(self, workflow: "LzyWorkflow") -> None:
         from lzy_amd.utils.metrics import timed
 
+        if workflow._snapshot is None:
+            return  # never started: nothing to drop/close
         with timed("lzy_wf_finish"):
             # never-consumed results may still be read after the block
             # exits (reference tutorial 3 prints an op result outside
@@ -720,6 +729,11 @@ class _DriverScheduler:
         from lzy_amd.config import get_config
 
         self._chain_enabled = bool(getattr(get_config(), "chain_dispatch", True))
+        # worker-death recovery: per-task re-dispatch budget (reference:
+        # scheduler re-allocation after VM death + storage-peer failover)
+        self._max_retries = int(getattr(get_config(), "task_retries", 1))
+        self.retry_budget: Dict[str, int] = {}
+        self.retrying: Set[str] = set()
         from lzy_amd.channels.transport import ipc_enabled
 
         # explicit ipc mode, or automatic when ranks outnumber GPUs:
@@ -776,9 +790,17 @@ class _DriverScheduler:
 
         pool = self.pool
         failed_tasks: Set[str] = set()
+        self._failed_tasks = failed_tasks
         while self.inflight > 0:
             rank, msg = pool.events.get()
             ev = msg.get("ev")
+            if ev in ("task_done", "task_failed"):
+                tid = msg["result"].task_id
+                if tid not in self.calls or tid not in self.task_dispatch_ts:
+                    # stale event: a prior batch's leftover, or a task
+                    # already terminally accounted (agent_error /
+                    # worker_lost) — must not touch inflight again
+                    continue
             if ev == "task_done":
                 result: TaskResult = msg["result"]
                 finished = self._on_done(rank, result)
@@ -817,33 +839,37 @@ class _DriverScheduler:
                 self.errors.append(
                     LzyExecutionError(f"agent rank {rank}: {msg['error']}")
                 )
+                tid = msg.get("task_id")
+                if tid is None or tid not in self.task_dispatch_ts:
+                    # serve-loop failure on a non-task command (transfer
+                    # staging etc.): the waiting task surfaces its own
+                    # failure later — decrementing inflight here would
+                    # underflow and leak its completion event into the
+                    # next batch's scheduler
+                    continue
+                if rank in self.outstanding:
+                    self.outstanding[rank] -= 1
+                if tid not in failed_tasks:
+                    failed_tasks.add(tid)
+                    self.journal.record(tid, "failed", msg["error"])
+                    for ct in dag.fail(tid):
+                        if ct in self.dispatched:
+                            failed_tasks.add(ct)
+                            self._poison_chained(ct, tid)
+                        else:
+                            self.journal.record(ct, "cancelled")
+                gang = self.gang_pending.get(tid)
+                if gang is not None:
+                    gang.discard(rank)
+                    if gang:
+                        continue  # other gang members still must report
+                    self.gang_pending.pop(tid, None)
+                    self._release_gang(tid)
                 self.inflight -= 1
+                self.task_dispatch_ts.pop(tid, None)
+                self.chained_waits.pop(tid, None)
             elif ev == "worker_lost":
-                # fail every task inflight on the dead rank; exclude the
-                # rank from further placement (reference: dead-VM cleanup)
-                self.outstanding.pop(rank, None)
-                dead = [
-                    tid for tid, rks in self.task_ranks.items()
-                    if rank in rks and self.task_dispatch_ts.get(tid) is not None
-                ]
-                for tid in dead:
-                    self.task_dispatch_ts.pop(tid, None)
-                    self.errors.append(LzyExecutionError(
-                        f"worker rank {rank} died while running "
-                        f"{self.calls[tid].callable_name}", task_id=tid,
-                    ))
-                    self.journal.record(tid, "failed", f"worker {rank} lost")
-                    if tid not in failed_tasks:
-                        failed_tasks.add(tid)
-                        for ct in dag.fail(tid):
-                            if ct in self.dispatched:
-                                # chained dependent inflight on a LIVE
-                                # rank: poison its waits now
-                                failed_tasks.add(ct)
-                                self._poison_chained(ct, tid)
-                            else:
-                                self.journal.record(ct, "cancelled")
-                    self.inflight -= 1
+                self._on_worker_lost(rank, dag, failed_tasks)
             # barrier_done etc. are routed via acks, not here
 
         if self.errors:
@@ -1110,6 +1136,133 @@ class _DriverScheduler:
             # tasks were already failed by the worker_lost sweep
             pass
 
+    # -- worker death --------------------------------------------------------
+
+    def _on_worker_lost(self, rank: int, dag, failed_tasks: Set[str]) -> None:
+        """A worker died (its control socket closed).  Exclude it from
+        placement, purge its data ownership, and re-dispatch its inflight
+        tasks onto surviving ranks when their inputs are recoverable
+        (reference failover: consumers re-pointed at the channel's storage
+        peer + scheduler re-allocation, SlotsService.java:191-240);
+        unrecoverable tasks fail the workflow as before.
+
+        Not recovered: gang tasks (their RCCL subgroup spans the dead
+        rank; surviving members' collectives cannot complete) and tasks
+        whose only input copy died with the rank and has no durable blob.
+        """
+        self.outstanding.pop(rank, None)
+        for m in self.meta.values():
+            if rank in m.owners:
+                m.owners.discard(rank)
+                # the cached handle may map the dead process's allocation
+                m.ipc_handle = None
+        self.transferred_now = {
+            (r, e) for (r, e) in self.transferred_now if r != rank
+        }
+        dead = [
+            tid for tid, rks in self.task_ranks.items()
+            if rank in rks and tid in self.task_dispatch_ts
+        ]
+        # retry eligibility, then input recovery to a fixpoint: a chained
+        # child's lost input is fine iff its producer re-runs in this
+        # same sweep — a demoted producer demotes the child too
+        self.retrying = {
+            tid for tid in dead
+            if self.outstanding
+            and self.retry_budget.get(tid, self._max_retries) > 0
+            and self.calls[tid].env.provisioning.effective_gpu_count <= 1
+        }
+        changed = True
+        while changed:
+            changed = False
+            for tid in list(self.retrying):
+                if not self._recover_inputs(self.calls[tid]):
+                    self.retrying.discard(tid)
+                    changed = True
+        for tid in dead:
+            self.task_dispatch_ts.pop(tid, None)
+            self.chained_waits.pop(tid, None)
+            gang = self.gang_pending.pop(tid, None)
+            if gang is not None:
+                self._release_gang(tid)
+            self.inflight -= 1
+            self.task_ranks.pop(tid, None)
+            if tid in self.retrying:
+                self.retry_budget[tid] = (
+                    self.retry_budget.get(tid, self._max_retries) - 1
+                )
+                self.dispatched.discard(tid)
+                self.journal.record(tid, "retry", f"worker {rank} lost")
+                METRICS.inc("lzy_task_retries")
+                _LOG.warning(
+                    "worker rank %d died; re-dispatching %s",
+                    rank, self.calls[tid].callable_name,
+                )
+            else:
+                self.errors.append(LzyExecutionError(
+                    f"worker rank {rank} died while running "
+                    f"{self.calls[tid].callable_name}", task_id=tid,
+                ))
+                self.journal.record(tid, "failed", f"worker {rank} lost")
+                if tid not in failed_tasks:
+                    failed_tasks.add(tid)
+                    for ct in dag.fail(tid):
+                        if ct in self.dispatched:
+                            # chained dependent inflight on a LIVE rank:
+                            # poison its waits now
+                            failed_tasks.add(ct)
+                            self._poison_chained(ct, tid)
+                        else:
+                            self.journal.record(ct, "cancelled")
+        # re-dispatch retried tasks whose deps are already satisfied; the
+        # rest re-dispatch when their (also-retried) producers complete,
+        # through the run loop's normal dag.complete path
+        for tid in sorted(self.retrying):
+            if tid in self.dispatched:
+                continue
+            if all(d in self.dag_completed for d in self.task_deps.get(tid, ())):
+                self._dispatch(tid)
+        self.retrying = set()
+
+    def _recover_inputs(self, call: "LzyCall") -> bool:
+        """Check every input of a to-be-retried task is reachable from a
+        surviving rank — restoring driver ownership from the durable tier
+        where needed.  Returns False when any input is gone for good."""
+        snap = self.workflow.snapshot
+        store = self.pool.agent.store
+        for eid in call.input_entry_ids():
+            prod = self.entry_producer.get(eid)
+            if prod is not None and prod in self.retrying:
+                continue  # will be re-produced by the retried producer
+            meta = self.meta.get(eid)
+            if meta is None:
+                # driver-captured arg: lives in the rank-0 store/snapshot
+                if store.has(eid) or snap.has_value(eid):
+                    continue
+                return False
+            if any(o in self.outstanding for o in meta.owners):
+                continue
+            if store.has(eid):
+                meta.owners.add(0)
+                continue
+            # durable tier: cache blobs / persisted entries reload on the
+            # driver, which becomes the new owner
+            try:
+                entry = snap.get_entry(eid)
+            except KeyError:
+                return False
+            if not snap.storage.blob_exists(entry.storage_uri):
+                return False
+            value = snap.load(eid)  # lands in the rank-0 store (shared dict)
+            fresh = describe_value(eid, value)
+            fresh.owners = {0}
+            if fresh.kind == KIND_BYTES:
+                data = pickle_value(value)
+                store.pickled[eid] = data
+                fresh.nbytes = len(data)
+            self.meta[eid] = fresh
+        return True
+
     def _try_chain(self) -> None:
         """Dispatch tasks whose producers are all inflight/complete on one
         rank, without waiting for their completion events."""
@@ -1209,3 +1362,7 @@ class _DriverScheduler:
             self.gang_pending.pop(result.task_id, None)
             self._release_gang(result.task_id)
         self.inflight -= 1
+        # terminal: dispatch_ts doubles as the liveness marker that the
+        # run loop's stale-event guard checks
+        self.task_dispatch_ts.pop(result.task_id, None)
+        self.chained_waits.pop(result.task_id, None)

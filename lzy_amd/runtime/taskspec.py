@@ -116,13 +116,16 @@ class WorkerStore:
                 self._cond.notify_all()
 
     def get(self, entry_id: str) -> Any:
-        if self.spill.is_spilled(entry_id):
-            value = self.spill.unspill(entry_id, self.values)
-            return value
-        value = self.values[entry_id]
+        # unspill returns None when the entry is not spilled — including
+        # when a concurrent get() unspilled it first, in which case the
+        # device tensor is already back in self.values (written under the
+        # spill lock): fall through instead of returning None.
+        value = self.spill.unspill(entry_id, self.values)
+        if value is None:
+            value = self.values[entry_id]
+            self.spill.track(entry_id, value)  # LRU touch
         if isinstance(value, _PoisonEntry):
             raise RuntimeError(f"input {entry_id} unavailable: {value.reason}")
-        self.spill.track(entry_id, value)  # LRU touch
         from lzy_amd.runtime.streams import STREAMS
 
         STREAMS.wait_value(entry_id, value)  # order after producing stream

@@ -152,8 +152,11 @@ class DefaultSnapshot:
 
     def try_get(self, entry_id: str) -> TryGetResult:
         if entry_id in self._values:
-            if self.spill.is_spilled(entry_id):
-                return TryGetResult(True, self.spill.unspill(entry_id, self._values))
+            # unspill returns None when not spilled (or when a concurrent
+            # unspill won the race and the device tensor is already back)
+            value = self.spill.unspill(entry_id, self._values)
+            if value is not None:
+                return TryGetResult(True, value)
             value = self._values[entry_id]
             self.spill.track(entry_id, value)  # LRU touch
             # cross-stream hand-off safety: the reader's stream waits on

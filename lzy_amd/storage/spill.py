@@ -170,8 +170,16 @@ class SpillManager:
             if ev is not None:
                 ev.synchronize()  # device copy may be freed only after D2H
             with self._lock:
+                # the _spilled insert and the store swap must be ONE
+                # atomic step: a concurrent get()/unspill between them
+                # would unspill (no-op on store) and then be overwritten
+                # by the host tensor below, stranding a CPU tensor in a
+                # device entry.  Also abort if the entry was dropped or
+                # replaced while the D2H was in flight.
+                if store_values.get(victim) is not value:
+                    continue
                 self._spilled[victim] = (host, value.device, ev)
-            store_values[victim] = host  # device tensor ref dropped -> freed
+                store_values[victim] = host  # device ref dropped -> freed
             nbytes = value.numel() * value.element_size()
             METRICS.inc("lzy_spill_out")
             METRICS.inc("lzy_spill_bytes", nbytes)
@@ -181,20 +189,31 @@ class SpillManager:
 
     def unspill(self, entry_id: str, store_values: Dict[str, Any]) -> Optional[torch.Tensor]:
         """Bring a spilled entry back to its device; returns the device
-        tensor (and re-registers it in the LRU)."""
+        tensor (and re-registers it in the LRU).  Returns None when the
+        entry is not spilled — including when a concurrent unspill won the
+        race, in which case ``store_values`` already holds the device
+        tensor by the time the lock is released here.
+
+        Fully locked (pop + H2D + store write): the H2D is a rare slow
+        path, and partial visibility (popped from _spilled but not yet
+        back in the store) would let maybe_spill/get interleave wrongly.
+        """
         with self._lock:
             hit = self._spilled.pop(entry_id, None)
-        if hit is None:
-            return None
-        host, device, _ = hit
-        t = self._mover.to_device(host, device)
-        # re-publish a producing event BEFORE publication (the H2D is
-        # ordered into the current stream by the mover)
-        from lzy_amd.runtime.streams import STREAMS
+            if hit is None:
+                return None
+            host, device, _ = hit
+            t = self._mover.to_device(host, device)
+            # re-publish a producing event BEFORE publication (the H2D is
+            # ordered into the current stream by the mover)
+            from lzy_amd.runtime.streams import STREAMS
 
-        if isinstance(t, torch.Tensor) and t.is_cuda:
-            STREAMS.record_output(entry_id, t)
-        store_values[entry_id] = t
-        self.track(entry_id, t)
+            if isinstance(t, torch.Tensor) and t.is_cuda:
+                STREAMS.record_output(entry_id, t)
+            store_values[entry_id] = t
+            # inline LRU touch (track() re-acquires this lock)
+            if getattr(t, "is_cuda", False):
+                self._lru.pop(entry_id, None)
+                self._lru[entry_id] = t.numel() * t.element_size()
         METRICS.inc("lzy_spill_in")
         return t

@@ -1,5 +1,13 @@
-"""Worker-death scenario: rank 1 hard-exits mid-op; the driver must
-detect it, fail the affected tasks, and raise instead of hanging."""
+"""Worker-death scenario: rank 1 hard-exits mid-op.
+
+Two modes, selected by LZY_TASK_RETRIES:
+  * retries > 0 (default): the driver detects the death, purges the dead
+    rank's ownership, and RE-DISPATCHES its inflight tasks onto surviving
+    ranks — the workflow COMPLETES (reference failover: storage-peer
+    re-pointing + scheduler re-allocation, SlotsService.java:191-240).
+  * retries = 0: recovery disabled — the driver must fail the affected
+    tasks and raise instead of hanging.
+"""
 import os
 import sys
 import time
@@ -16,7 +24,7 @@ def maybe_die(i: int) -> int:
     if os.environ.get("RANK") == "1":
         os._exit(7)
     time.sleep(0.2)
-    return i
+    return i * 10
 
 
 def main() -> None:
@@ -25,8 +33,9 @@ def main() -> None:
     try:
         with lzy.workflow("death"):
             rs = [maybe_die(i) for i in range(6)]
-            [int(r) for r in rs]
-        print("DEATH-NOT-DETECTED", flush=True)
+            vals = [int(r) for r in rs]
+        assert vals == [i * 10 for i in range(6)], vals
+        print("DEATH-RECOVERED", flush=True)
     except LzyExecutionError as e:
         print(f"DEATH-DETECTED: {e}", flush=True)
     os._exit(0)

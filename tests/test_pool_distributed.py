@@ -64,7 +64,7 @@ def test_pool_single_rank_inprocess(tmp_path, monkeypatch):
     assert "SINGLE-OK" in res.stdout
 
 
-def test_worker_death_detected(tmp_path):
+def _run_death_scenario(tmp_path, extra_env):
     # launch ranks manually: torchrun would kill the driver the moment
     # rank 1 exits, hiding exactly the recovery we want to observe
     import socket
@@ -78,6 +78,7 @@ def test_worker_death_detected(tmp_path):
     base_env.update(
         MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port), WORLD_SIZE="2"
     )
+    base_env.update(extra_env)
     procs = []
     for rank in (0, 1):
         env = dict(base_env)
@@ -91,8 +92,21 @@ def test_worker_death_detected(tmp_path):
     out0, err0 = procs[0].communicate(timeout=120)
     procs[1].wait(timeout=30)
     assert procs[1].returncode == 7  # the injected hard exit
+    return out0, err0
+
+
+def test_worker_death_recovers(tmp_path):
+    """Default: inflight tasks of the dead rank re-dispatch to survivors
+    and the workflow COMPLETES (reference: SlotsService failover)."""
+    out0, err0 = _run_death_scenario(tmp_path, {})
+    assert "DEATH-RECOVERED" in out0, out0[-3000:] + err0[-2000:]
+
+
+def test_worker_death_detected_without_retries(tmp_path):
+    """task_retries=0: recovery off — the failure must surface, not hang."""
+    out0, err0 = _run_death_scenario(tmp_path, {"LZY_TASK_RETRIES": "0"})
     assert "DEATH-DETECTED" in out0, out0[-3000:] + err0[-2000:]
-    assert "DEATH-NOT-DETECTED" not in out0
+    assert "DEATH-RECOVERED" not in out0
 
 
 def test_chunked_transport_offset_resume(tmp_path):
