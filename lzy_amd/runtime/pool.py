@@ -107,6 +107,10 @@ class WorkerAgent:
         self._groups: Dict[str, Any] = {}
         self._exec_q: "queue.Queue[dict]" = queue.Queue()
         self._shutdown = False
+        # StopGraph support: task ids cancelled mid-flight (reference:
+        # AbortExecution stopGraphs, workflow-service.proto StopGraph).
+        # Queued tasks check this when an executor thread picks them up.
+        self._cancelled: Dict[str, str] = {}
         # small task pool per rank: independent tasks of one rank overlap
         # (each executor thread gets its own HIP stream via
         # runtime/streams.py).  Gang tasks are serialized DRIVER-side
@@ -215,6 +219,10 @@ class WorkerAgent:
         elif cmd == "load_serializers":
             self.serializers.load_user_serializers(msg["payload"])
             self.ctrl.send_event({"ev": "ack", "tag": msg["tag"], "rank": self.rank})
+        elif cmd == "cancel_tasks":
+            reason = msg.get("reason", "graph stopped")
+            for tid in msg["ids"]:
+                self._cancelled[tid] = reason
         elif cmd == "poison":
             for eid in msg["entries"]:
                 self.store.poison(eid, msg.get("reason", "producer failed"))
@@ -223,6 +231,7 @@ class WorkerAgent:
                 self.store.drop(eid)
         elif cmd == "clear_store":
             self.store.clear()
+            self._cancelled.clear()
         elif cmd == "shutdown":
             self._shutdown = True
             self._exec_q.put({"cmd": "_stop"})
@@ -286,6 +295,16 @@ class WorkerAgent:
 
     def _run_guarded(self, msg: dict) -> None:
         cmd = msg["cmd"]
+        spec = msg.get("spec")
+        if spec is not None and spec.task_id in self._cancelled:
+            # cancelled while queued: skip the op entirely (the running
+            # ones can't be killed — this is what bounds abort latency
+            # to ONE op duration instead of the whole queue)
+            self.ctrl.send_event({
+                "ev": "task_cancelled", "task_id": spec.task_id,
+                "rank": self.rank,
+            })
+            return
         try:
             if cmd == "task":
                 self._run_task(msg)
@@ -553,6 +572,7 @@ class GpuPoolRuntime(Runtime):
         # different thread's active flight.
         self._flight = threading.Lock()
         self._flight_owner: Optional[str] = None
+        self._active_sched: Optional["_DriverScheduler"] = None
 
     @property
     def pool(self) -> GpuPool:
@@ -614,7 +634,11 @@ class GpuPoolRuntime(Runtime):
 
         with timed("lzy_wf_exec"):
             sched = _DriverScheduler(self.pool, workflow, calls, self._journal)
-            sched.run()
+            self._active_sched = sched
+            try:
+                sched.run()
+            finally:
+                self._active_sched = None
 
     def finish(self, workflow: "LzyWorkflow") -> None:
         from lzy_amd.utils.metrics import timed
@@ -648,6 +672,16 @@ class GpuPoolRuntime(Runtime):
                 self._journal = None
 
     def abort(self, workflow: "LzyWorkflow") -> None:
+        """Stop the graph, then finish.  When a scheduler batch is live
+        (another thread is blocked in its barrier), request a mid-flight
+        StopGraph and wait for the drain before releasing resources
+        (reference: AbortExecution stops graphs first,
+        AbortExecution.java:18; StopGraph workflow-service.proto:12-26).
+        """
+        sched = self._active_sched
+        if sched is not None and sched.workflow is workflow:
+            sched.request_stop("workflow aborted by client")
+            sched._done_evt.wait(timeout=60.0)
         self.finish(workflow)
 
     def _drop_workflow_entries(self, workflow: "LzyWorkflow",
@@ -734,6 +768,10 @@ class _DriverScheduler:
         self._max_retries = int(getattr(get_config(), "task_retries", 1))
         self.retry_budget: Dict[str, int] = {}
         self.retrying: Set[str] = set()
+        # mid-flight StopGraph: reason once set; done event lets an
+        # aborting thread wait for the drain to finish
+        self.stopping: Optional[str] = None
+        self._done_evt = threading.Event()
         from lzy_amd.channels.transport import ipc_enabled
 
         # explicit ipc mode, or automatic when ranks outnumber GPUs:
@@ -788,11 +826,85 @@ class _DriverScheduler:
             self._dispatch(tid)
         self._try_chain()
 
-        pool = self.pool
         failed_tasks: Set[str] = set()
         self._failed_tasks = failed_tasks
+        try:
+            self._event_loop(dag, failed_tasks)
+        except (KeyboardInterrupt, SystemExit):
+            # Ctrl-C mid-barrier: stop the graph (cancel queued tasks on
+            # every rank, poison blocked waits) and drain what is truly
+            # running, bounded — then let the interrupt propagate
+            self._initiate_stop(dag, "interrupted (Ctrl-C)")
+            try:
+                self._event_loop(
+                    dag, failed_tasks, deadline=time.monotonic() + 30.0
+                )
+            except BaseException:  # noqa: BLE001 - interrupt wins
+                pass
+            raise
+        finally:
+            self._done_evt.set()
+
+        if self.stopping is not None:
+            from lzy_amd.exceptions import WorkflowAbortedError
+
+            raise WorkflowAbortedError(
+                f"graph stopped: {self.stopping}"
+            )
+        if self.errors:
+            raise self.errors[0]
+
+    def request_stop(self, reason: str) -> None:
+        """Thread-safe mid-flight StopGraph (reference: StopGraph RPC,
+        workflow-service.proto:12-26; AbortExecution.java:18).  Wakes the
+        event loop; it cancels everything not yet running and drains."""
+        self.pool.events.put(
+            (-1, {"ev": "stop", "reason": reason, "sched": id(self)})
+        )
+
+    def _initiate_stop(self, dag, reason: str) -> None:
+        if self.stopping is not None:
+            return
+        self.stopping = reason
+        pool = self.pool
+        self.deferred_gangs = []
+        for tid in self.calls:
+            if tid not in self.dispatched:
+                self.journal.record(tid, "cancelled", reason)
+        live = list(self.task_dispatch_ts)
+        if live:
+            pool.driver_ctrl.broadcast(
+                {"cmd": "cancel_tasks", "ids": live, "reason": reason}
+            )
+        # blocked settles (chained consumers waiting on producers that
+        # will now never run) must fail fast, not time out
+        for tid, (rank, eids) in list(self.chained_waits.items()):
+            try:
+                pool.driver_ctrl.send(rank, {
+                    "cmd": "poison", "entries": eids, "reason": reason,
+                })
+            except (OSError, KeyError, BrokenPipeError):
+                pass
+        _LOG.warning("graph stop initiated: %s (%d tasks inflight)",
+                     reason, self.inflight)
+
+    def _event_loop(self, dag, failed_tasks: Set[str],
+                    deadline: Optional[float] = None) -> None:
+        pool = self.pool
         while self.inflight > 0:
-            rank, msg = pool.events.get()
+            if deadline is None:
+                rank, msg = pool.events.get()
+            else:
+                try:
+                    rank, msg = pool.events.get(
+                        timeout=max(0.05, deadline - time.monotonic())
+                    )
+                except queue.Empty:
+                    _LOG.warning(
+                        "stop drain abandoned with %d tasks inflight",
+                        self.inflight,
+                    )
+                    return
             ev = msg.get("ev")
             if ev in ("task_done", "task_failed"):
                 tid = msg["result"].task_id
@@ -868,17 +980,39 @@ class _DriverScheduler:
                 self.inflight -= 1
                 self.task_dispatch_ts.pop(tid, None)
                 self.chained_waits.pop(tid, None)
+            elif ev == "task_cancelled":
+                tid = msg["task_id"]
+                if tid not in self.task_dispatch_ts:
+                    continue
+                if rank in self.outstanding:
+                    self.outstanding[rank] -= 1
+                gang = self.gang_pending.get(tid)
+                if gang is not None:
+                    gang.discard(rank)
+                    if gang:
+                        continue
+                    self.gang_pending.pop(tid, None)
+                    self._release_gang(tid)
+                self.inflight -= 1
+                self.task_dispatch_ts.pop(tid, None)
+                self.chained_waits.pop(tid, None)
+                self.journal.record(tid, "cancelled", self.stopping or "")
+            elif ev == "stop":
+                if msg.get("sched") == id(self):
+                    self._initiate_stop(dag, msg.get("reason", "stopped"))
+                # a foreign scheduler's stop (stale from a prior batch)
+                # is dropped
             elif ev == "worker_lost":
                 self._on_worker_lost(rank, dag, failed_tasks)
             # barrier_done etc. are routed via acks, not here
-
-        if self.errors:
-            raise self.errors[0]
 
     # -- dispatch ------------------------------------------------------------
 
     def _dispatch(self, task_id: str, chain_rank: Optional[int] = None,
                   pending_local: Optional[Set[str]] = None) -> None:
+        if self.stopping is not None:
+            self.journal.record(task_id, "cancelled", self.stopping)
+            return
         call = self.calls[task_id]
         gpu_count = call.env.provisioning.effective_gpu_count
         pool = self.pool
@@ -1266,7 +1400,7 @@ class _DriverScheduler:
     def _try_chain(self) -> None:
         """Dispatch tasks whose producers are all inflight/complete on one
         rank, without waiting for their completion events."""
-        if not self._chain_enabled:
+        if not self._chain_enabled or self.stopping is not None:
             return
         for tid, call in self.calls.items():
             if tid in self.dispatched:

@@ -109,6 +109,14 @@ def test_worker_death_detected_without_retries(tmp_path):
     assert "DEATH-RECOVERED" not in out0
 
 
+def test_stop_graph_mid_flight(tmp_path):
+    """Abort of a slow fan-out mid-barrier: queued tasks cancelled on all
+    ranks, barrier raises promptly, pool stays usable (StopGraph)."""
+    r = _run_distributed("tests/pool_script_stop.py", 2, tmp_path)
+    assert r.returncode == 0, r.stdout[-3000:] + r.stderr[-3000:]
+    assert "STOP-OK" in r.stdout
+
+
 def test_chunked_transport_offset_resume(tmp_path):
     r = _run_distributed("tests/chunked_transport_script.py", 2, tmp_path)
     assert r.returncode == 0, r.stdout[-3000:] + r.stderr[-3000:]
