@@ -113,9 +113,9 @@ class WorkerAgent:
         self._cancelled: Dict[str, str] = {}
         # small task pool per rank: independent tasks of one rank overlap
         # (each executor thread gets its own HIP stream via
-        # runtime/streams.py).  Gang tasks are serialized DRIVER-side
-        # (one gang inflight at a time) so collective order can never
-        # diverge across ranks.
+        # runtime/streams.py).  Gang tasks are admitted DRIVER-side only
+        # onto rank sets disjoint from every inflight gang, so collective
+        # order can never diverge on any rank.
         from concurrent.futures import ThreadPoolExecutor
 
         from lzy_amd.config import get_config
@@ -746,9 +746,12 @@ class _DriverScheduler:
         # with intra-rank task concurrency, every later consumer task on
         # that rank must also wait for the settle (taskspec.wait_present)
         self.transferred_now: Set[Tuple[int, str]] = set()
-        # collectives must start in the same order on every rank: at most
-        # one gang task inflight at a time, the rest queue here
-        self.gang_inflight: Optional[str] = None
+        # collectives must start in the same order on every rank.  Gangs
+        # with DISJOINT rank sets cannot interleave collectives on any
+        # rank, so they run concurrently (reference: per-task LRO
+        # concurrency, ExecuteTaskAction.java:44); a gang that cannot
+        # get a disjoint slot queues here.  Group creation stays
+        # driver-sequenced (ensure_group broadcast + all-rank acks).
         self.deferred_gangs: List[str] = []
         # chain dispatch: a task whose producers are all inflight on ONE
         # rank is dispatched there immediately (worker FIFO + store
@@ -1029,13 +1032,15 @@ class _DriverScheduler:
             ranks = [chain_rank]
             gang = None
         elif gpu_count > 1:
-            if self.gang_inflight is not None:
+            busy_gang_ranks: Set[int] = set()
+            for t in self.gang_pending:
+                busy_gang_ranks.update(self.task_ranks.get(t, ()))
+            ranks = self._pick_gang(gpu_count, exclude=busy_gang_ranks)
+            if ranks is None:
                 self.deferred_gangs.append(task_id)
                 return
-            ranks = self._pick_gang(gpu_count)
             tag = pool.ensure_group(ranks)
             gang = {"ranks": ranks, "tag": tag}
-            self.gang_inflight = task_id
         else:
             ranks = [self._pick_rank(call)]
             gang = None
@@ -1204,10 +1209,15 @@ class _DriverScheduler:
         self.pool.rr_counter = getattr(self.pool, "rr_counter", -1) + 1
         return cands[self.pool.rr_counter % len(cands)]
 
-    def _pick_gang(self, k: int) -> List[int]:
-        ranks = sorted(
-            self.outstanding, key=lambda r: (self.outstanding[r], r)
-        )[:k]
+    def _pick_gang(self, k: int,
+                   exclude: Optional[Set[int]] = None) -> Optional[List[int]]:
+        """Least-loaded k ranks disjoint from every inflight gang;
+        None when no disjoint slot exists right now."""
+        exclude = exclude or set()
+        cands = [r for r in self.outstanding if r not in exclude]
+        if len(cands) < k:
+            return None
+        ranks = sorted(cands, key=lambda r: (self.outstanding[r], r))[:k]
         return sorted(ranks)
 
     # -- completion ----------------------------------------------------------
@@ -1437,10 +1447,13 @@ class _DriverScheduler:
             METRICS.inc("lzy_chain_dispatches")
 
     def _release_gang(self, task_id: str) -> None:
-        if self.gang_inflight == task_id:
-            self.gang_inflight = None
-            if self.deferred_gangs:
-                self._dispatch(self.deferred_gangs.pop(0))
+        """A gang finished: its ranks are free — re-try every deferred
+        gang (more than one may now fit disjoint slots)."""
+        if self.deferred_gangs:
+            pending = self.deferred_gangs
+            self.deferred_gangs = []  # _dispatch re-defers what still can't fit
+            for tid in pending:
+                self._dispatch(tid)
 
     def _record_outputs(self, call: "LzyCall", rank: int, result: TaskResult) -> None:
         snap = self.workflow.snapshot

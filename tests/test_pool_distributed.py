@@ -109,6 +109,14 @@ def test_worker_death_detected_without_retries(tmp_path):
     assert "DEATH-RECOVERED" not in out0
 
 
+def test_concurrent_gangs_world8(tmp_path):
+    """Two+ disjoint gpu_count=2 gangs and singles run simultaneously
+    (driver-sequenced group creation, disjoint-rank admission)."""
+    r = _run_distributed("tests/pool_script_gangs.py", 8, tmp_path, timeout=300)
+    assert r.returncode == 0, r.stdout[-3000:] + r.stderr[-3000:]
+    assert "GANGS-OK" in r.stdout
+
+
 def test_stop_graph_mid_flight(tmp_path):
     """Abort of a slow fan-out mid-barrier: queued tasks cancelled on all
     ranks, barrier raises promptly, pool stays usable (StopGraph)."""
