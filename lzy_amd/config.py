@@ -65,6 +65,7 @@ class Config:
     spill_threshold_frac: float = 0.85
     # logs / metrics
     log_archive: bool = True           # archive per-op std logs to durable tier
+    log_stream_period_s: float = 0.25  # remote-rank live log flush period
     metrics_port: int = 0              # >0 -> serve /metrics on this port
     status_port: int = 0               # >0 -> serve status endpoint on this port
     # client-visible identity (reference: lzy_auth)

@@ -130,6 +130,19 @@ class WorkerAgent:
             target=self._executor, daemon=True, name=f"lzy-exec-r{rank}"
         )
         self._exec_thread.start()
+        # live log streaming from ranks > 0 (reference: worker→Kafka→
+        # client ReadStdSlots stream while the op runs): a flusher ships
+        # newline-bounded buffer deltas to the driver periodically
+        self._livelogs: Dict[str, list] = {}  # tid -> [name, out, err, so, se]
+        self._livelog_lock = threading.Lock()
+        if rank != 0:
+            self._log_period = float(
+                getattr(get_config(), "log_stream_period_s", 0.25)
+            )
+            threading.Thread(
+                target=self._livelog_flusher, daemon=True,
+                name=f"lzy-logflush-r{rank}",
+            ).start()
         OpLogCapture.instance().install()
 
     # -- serve loop ---------------------------------------------------------
@@ -357,6 +370,68 @@ class WorkerAgent:
             STREAMS.record_output(eid, value)
             self.store.put(eid, value, pickled=pickled)
 
+    # -- live log streaming --------------------------------------------------
+
+    def _live_sink_for(self, spec: "TaskSpec"):
+        """Registration hook handed to run_taskspec on ranks > 0."""
+        if self.rank == 0:
+            return None  # echoed live on the shared console already
+
+        def register(out_buf, err_buf):
+            tid = spec.task_id
+            with self._livelog_lock:
+                self._livelogs[tid] = [spec.name, out_buf, err_buf, 0, 0]
+
+            def unregister():
+                self._flush_livelog(tid, final=True)
+                with self._livelog_lock:
+                    self._livelogs.pop(tid, None)
+
+            return unregister
+
+        return register
+
+    def _flush_livelog(self, tid: str, final: bool = False) -> None:
+        with self._livelog_lock:
+            rec = self._livelogs.get(tid)
+            if rec is None:
+                return
+            name, out_buf, err_buf, so, se = rec
+
+        def delta(buf, shipped):
+            s = buf.getvalue()[shipped:]
+            if not final:
+                # hold partial lines: the driver prints whole lines
+                cut = s.rfind("\n") + 1
+                s = s[:cut]
+            return s
+
+        d_out = delta(out_buf, so)
+        d_err = delta(err_buf, se)
+        if not d_out and not d_err:
+            return
+        with self._livelog_lock:
+            rec = self._livelogs.get(tid)
+            if rec is None:
+                return
+            rec[3] += len(d_out)
+            rec[4] += len(d_err)
+        try:
+            self.ctrl.send_event({
+                "ev": "log_chunk", "task_id": tid, "name": name,
+                "rank": self.rank, "out": d_out, "err": d_err,
+            })
+        except (OSError, BrokenPipeError):
+            pass
+
+    def _livelog_flusher(self) -> None:
+        while not self._shutdown:
+            time.sleep(self._log_period)
+            with self._livelog_lock:
+                tids = list(self._livelogs)
+            for tid in tids:
+                self._flush_livelog(tid)
+
     def _run_task(self, msg: dict) -> None:
         spec: TaskSpec = msg["spec"]
         self._settle(spec.wait_entries)
@@ -369,10 +444,11 @@ class WorkerAgent:
         result = run_taskspec(
             spec, self.store, self.serializers, self.storage,
             gang_group=gang_group,
-            # rank 0 echoes op logs on its own console; other ranks ship
-            # them in the TaskResult and the driver tails them (the
-            # console belongs to the client)
+            # rank 0 echoes op logs on its own console; other ranks
+            # stream them live (log_chunk events) and ship the full
+            # capture in the TaskResult (the console belongs to the client)
             echo_logs=(self.rank == 0),
+            live_sink=self._live_sink_for(spec),
         )
         self.ctrl.send_event(
             {
@@ -775,6 +851,8 @@ class _DriverScheduler:
         # aborting thread wait for the drain to finish
         self.stopping: Optional[str] = None
         self._done_evt = threading.Event()
+        # (task_id, rank) -> [out_chars, err_chars] already streamed live
+        self.streamed_logs: Dict[Tuple[str, int], list] = {}
         from lzy_amd.channels.transport import ipc_enabled
 
         # explicit ipc mode, or automatic when ranks outnumber GPUs:
@@ -983,6 +1061,8 @@ class _DriverScheduler:
                 self.inflight -= 1
                 self.task_dispatch_ts.pop(tid, None)
                 self.chained_waits.pop(tid, None)
+            elif ev == "log_chunk":
+                self._print_log_chunk(msg)
             elif ev == "task_cancelled":
                 tid = msg["task_id"]
                 if tid not in self.task_dispatch_ts:
@@ -1222,17 +1302,36 @@ class _DriverScheduler:
 
     # -- completion ----------------------------------------------------------
 
+    def _print_log_chunk(self, msg: dict) -> None:
+        """A running remote op's incremental std-logs (reference:
+        ReadStdSlots server-stream fed live from Kafka,
+        KafkaLogsListeners.java:35): print now, remember how much was
+        streamed so the completion-time tail prints only the rest."""
+        name = msg["name"]
+        out, err = msg.get("out", ""), msg.get("err", "")
+        for line in out.splitlines():
+            print(f"[LZY-{name}] {line}", flush=True)
+        for line in err.splitlines():
+            print(f"[LZY-{name}] {line}", file=sys.stderr, flush=True)
+        shipped = self.streamed_logs.setdefault(
+            (msg["task_id"], msg["rank"]), [0, 0]
+        )
+        shipped[0] += len(out)
+        shipped[1] += len(err)
+
     def _tail_logs(self, rank: int, result: TaskResult) -> None:
         """Print a remote rank's captured op logs on the client console
-        (reference: ReadStdSlots live tail, runtime.py:283-301)."""
+        (reference: ReadStdSlots live tail, runtime.py:283-301).  Lines
+        already streamed live via log_chunk events are skipped."""
         if rank == 0:
             return  # echoed live on the shared console already
         name = self.calls[result.task_id].callable_name
-        if result.logs_out:
-            for line in result.logs_out.splitlines():
+        so, se = self.streamed_logs.pop((result.task_id, rank), (0, 0))
+        if result.logs_out[so:]:
+            for line in result.logs_out[so:].splitlines():
                 print(f"[LZY-{name}] {line}", flush=True)
-        if result.logs_err:
-            for line in result.logs_err.splitlines():
+        if result.logs_err[se:]:
+            for line in result.logs_err[se:].splitlines():
                 print(f"[LZY-{name}] {line}", file=sys.stderr, flush=True)
 
     def _on_done(self, rank: int, result: TaskResult) -> bool:

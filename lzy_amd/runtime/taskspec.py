@@ -158,6 +158,7 @@ def run_taskspec(
     storage,
     gang_group=None,
     echo_logs: bool = True,
+    live_sink=None,
 ) -> TaskResult:
     """Execute one TaskSpec against the worker store (worker innermost loop)."""
     from lzy_amd.channels.transport import describe_value, pickle_value, unpickle_value
@@ -209,6 +210,10 @@ def run_taskspec(
 
     capture = OpLogCapture.instance()
     out_buf, err_buf = capture.route_current_thread(spec.name, echo=echo_logs)
+    # live log streaming (reference: worker→Kafka→client ReadStdSlots
+    # stream while the op RUNS, KafkaLogsListeners.java:35): the agent's
+    # flusher ships buffer deltas to the driver until unregistered
+    unregister_live = live_sink(out_buf, err_buf) if live_sink else None
     old_env: Dict[str, Optional[str]] = {}
     # subprocesses spawned BY the op (user DDP launchers etc.) inherit
     # this marker and must not try to become pool drivers (reference:
@@ -255,6 +260,8 @@ def run_taskspec(
         tr.logs_err = err_buf.getvalue()[:_LOG_CAP]
         return tr
     finally:
+        if unregister_live is not None:
+            unregister_live()  # final flush precedes the completion event
         _set_in_op_execution(False)
         if spec.gang is not None:
             from lzy_amd.runtime.context import _set_op_context

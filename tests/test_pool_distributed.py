@@ -109,6 +109,14 @@ def test_worker_death_detected_without_retries(tmp_path):
     assert "DEATH-RECOVERED" not in out0
 
 
+def test_live_remote_log_tail(tmp_path):
+    """A remote rank's print appears on the driver console BEFORE the op
+    finishes (live ReadStdSlots-style streaming, not completion-time)."""
+    r = _run_distributed("tests/pool_script_livelogs.py", 2, tmp_path)
+    assert r.returncode == 0, r.stdout[-3000:] + r.stderr[-3000:]
+    assert "LIVELOGS-OK" in r.stdout
+
+
 def test_concurrent_gangs_world8(tmp_path):
     """Two+ disjoint gpu_count=2 gangs and singles run simultaneously
     (driver-sequenced group creation, disjoint-rank admission)."""
