@@ -156,8 +156,14 @@ def checksum(t: torch.Tensor) -> int:
     return int(_hash_bytes(t.float().numpy().tobytes()[:1 << 20]), 16)
 
 
-@op
+@op(pair_reduce=(0.5, 0.5))
 def merge(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    # pair_reduce declares the semantics (0.5*a + 0.5*b), letting the
+    # pool fold the whole merge tree into ONE chunk-streamed reduction
+    # plan over xGMI (channels/treeplan.py): a chunk combined at level k
+    # is forwarded to level k+1 immediately, so tree latency collapses
+    # from log2(N) transfers to ~one.  This body remains the semantic
+    # reference (LocalRuntime / same-rank / fallback paths).
     if a.is_cuda and b.is_cuda:
         from lzy_amd.ops import axpby
 

@@ -39,6 +39,7 @@ class LzyCall:
         version: str = "0.0",
         cache: bool = False,
         lazy_arguments: bool = False,
+        pair_reduce=None,
     ) -> None:
         self.id = str(uuid.uuid4())
         self.workflow = workflow
@@ -49,6 +50,9 @@ class LzyCall:
         self.version = version
         self.cache = cache
         self.lazy_arguments = lazy_arguments
+        # (alpha, beta) for ops declared as pairwise tensor reductions —
+        # lets the pool runtime fold merge trees into streamed plans
+        self.pair_reduce = pair_reduce
 
         snapshot = workflow.snapshot
         name = signature.qualname
@@ -107,6 +111,7 @@ class LazyCallWrapper(WithEnvironmentMixin):
         version: str = "0.0",
         cache: bool = False,
         lazy_arguments: bool = False,
+        pair_reduce=None,
     ) -> None:
         self.function = function
         self.output_types = tuple(output_types)
@@ -115,6 +120,7 @@ class LazyCallWrapper(WithEnvironmentMixin):
         self.version = version
         self.cache = cache
         self.lazy_arguments = lazy_arguments
+        self.pair_reduce = pair_reduce
         # look like the wrapped function
         self.__name__ = getattr(function, "__name__", "op")
         self.__qualname__ = getattr(function, "__qualname__", self.__name__)
@@ -129,6 +135,7 @@ class LazyCallWrapper(WithEnvironmentMixin):
             version=self.version,
             cache=self.cache,
             lazy_arguments=self.lazy_arguments,
+            pair_reduce=self.pair_reduce,
         )
         merged.update(kwargs)
         return LazyCallWrapper(**merged)
@@ -152,6 +159,7 @@ class LazyCallWrapper(WithEnvironmentMixin):
             version=self.version,
             cache=self.cache,
             lazy_arguments=self.lazy_arguments,
+            pair_reduce=self.pair_reduce,
         )
         workflow.register_call(call)
 

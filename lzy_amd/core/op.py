@@ -26,12 +26,20 @@ def op(
     gpu_type: Optional[str] = None,
     cpu_count: Optional[int] = None,
     ram_size_gb: Optional[int] = None,
+    pair_reduce=None,
 ):
     """Make a function lazily executable inside a workflow.
 
     ``gpu_count``/``gpu_type``/``cpu_count``/``ram_size_gb`` are provisioning
     shortcuts (reference routes them through env.provisioning; we accept
     them directly too because GPU placement is the common case here).
+
+    ``pair_reduce=(alpha, beta)`` declares the op computes the elementwise
+    ``alpha*a + beta*b`` of its two same-shape tensor arguments.  The pool
+    runtime may then fold trees of such ops into chunk-streamed reduction
+    plans over xGMI (channels/treeplan.py) — the function body remains the
+    semantic reference and still runs on the LocalRuntime / same-rank /
+    fallback paths.
     """
 
     def deco(f: Callable) -> LazyCallWrapper:
@@ -66,6 +74,7 @@ def op(
             version=version,
             cache=cache,
             lazy_arguments=lazy_arguments,
+            pair_reduce=tuple(pair_reduce) if pair_reduce is not None else None,
         )
 
     if func is None:

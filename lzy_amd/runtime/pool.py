@@ -93,10 +93,12 @@ def _storage_root() -> str:
 class WorkerAgent:
     """Per-rank serve loop + executor thread (every rank, driver included)."""
 
-    def __init__(self, rank: int, world: int, address: str, pg_data, device):
+    def __init__(self, rank: int, world: int, address: str, pg_data, device,
+                 pg_stream=None):
         self.rank = rank
         self.world = world
         self.device = device
+        self.pg_stream = pg_stream
         self.store = WorkerStore(device=device)
         self.serializers = LzySerializerRegistry()
         self.storage = FsStorageClient()
@@ -221,6 +223,8 @@ class WorkerAgent:
             )
         elif cmd == "settle":
             self._exec_q.put(msg)
+        elif cmd == "stream_plan":
+            self._exec_q.put(msg)
         elif cmd == "new_group":
             ranks = msg["ranks"]
             tag = msg["tag"]
@@ -265,7 +269,7 @@ class WorkerAgent:
             if cmd == "_stop":
                 self._task_pool.shutdown(wait=False)
                 return
-            if cmd in ("task", "settle"):
+            if cmd in ("task", "settle", "stream_plan"):
                 fut = self._task_pool.submit(self._run_guarded, msg)
                 with self._futs_lock:
                     self._futs.add(fut)
@@ -321,6 +325,8 @@ class WorkerAgent:
         try:
             if cmd == "task":
                 self._run_task(msg)
+            elif cmd == "stream_plan":
+                self._run_stream_plan(msg)
             else:
                 self._settle(msg["entries"])
                 self.ctrl.send_event(
@@ -329,6 +335,36 @@ class WorkerAgent:
         except BaseException as e:  # noqa: BLE001 - agent must survive
             _LOG.exception("agent r%d failed handling %s", self.rank, cmd)
             self._report_error(msg, e)
+
+    def _run_stream_plan(self, msg: dict) -> None:
+        """Execute this rank's schedule of a streamed merge-tree plan
+        (channels/treeplan.py); p2p runs on the dedicated pg_stream."""
+        from lzy_amd.channels.treeplan import run_stream_plan
+
+        plan = msg["plan"]
+        if plan["plan_id"] in self._cancelled:
+            self.ctrl.send_event({
+                "ev": "plan_failed", "plan_id": plan["plan_id"],
+                "rank": self.rank, "error": "cancelled",
+            })
+            return
+        try:
+            results = run_stream_plan(
+                plan, msg["steps"], self.store,
+                self.pg_stream, self.device,
+                wait_timeout=msg.get("wait_timeout", 120.0),
+            )
+        except BaseException as e:  # noqa: BLE001
+            _LOG.exception("agent r%d stream plan failed", self.rank)
+            self.ctrl.send_event({
+                "ev": "plan_failed", "plan_id": plan["plan_id"],
+                "rank": self.rank, "error": f"{type(e).__name__}: {e}",
+            })
+            return
+        self.ctrl.send_event({
+            "ev": "plan_done", "plan_id": plan["plan_id"],
+            "rank": self.rank, "outputs": results,
+        })
 
     def _report_error(self, msg: dict, e: BaseException) -> None:
         spec = msg.get("spec")
@@ -516,6 +552,12 @@ class GpuPool:
                     timeout=_dt.timedelta(seconds=180),
                 )
             pg_data = dist.new_group()  # dedicated transfer group
+            # streamed tree plans issue p2p from EXECUTOR threads; their
+            # own group keeps that traffic's issue order independent of
+            # the serve-loop-issued transfers on pg_data (separate comm =
+            # separate match space).  Created here so every rank joins
+            # the (eager, ncclCommSplit under device_id) creation.
+            self.pg_stream = dist.new_group()
 
         if self.is_driver:
             self.driver_ctrl = DriverControl(self.world, self._on_event)
@@ -529,7 +571,10 @@ class GpuPool:
         if self.world > 1:
             address = broadcast_address(address)
 
-        self.agent = WorkerAgent(self.rank, self.world, address, pg_data, self.device)
+        self.agent = WorkerAgent(
+            self.rank, self.world, address, pg_data, self.device,
+            pg_stream=getattr(self, "pg_stream", None),
+        )
 
         if self.is_driver:
             accept_thread.join()
@@ -853,6 +898,17 @@ class _DriverScheduler:
         self._done_evt = threading.Event()
         # (task_id, rank) -> [out_chars, err_chars] already streamed live
         self.streamed_logs: Dict[Tuple[str, int], list] = {}
+        # streamed merge-tree plans (channels/treeplan.py): connected
+        # components of pair_reduce tasks fold into one chunk-pipelined
+        # multi-rank reduction instead of log2(N) transfer-then-combine
+        # levels.  One plan active pool-wide (per-pair p2p issue order).
+        self.stream_components: Dict[str, int] = {}   # tid -> comp idx
+        self.components: Dict[int, dict] = {}
+        self.active_plan: Optional[dict] = None
+        self.deferred_plans: List[int] = []
+        self._stream_chunk = int(
+            getattr(get_config(), "stream_chunk_mb", 64)
+        ) << 20
         from lzy_amd.channels.transport import ipc_enabled
 
         # explicit ipc mode, or automatic when ranks outnumber GPUs:
@@ -865,6 +921,35 @@ class _DriverScheduler:
             and agent_tr is not None
             and not agent_tr._cuda_p2p
         )
+        self._stream_merge_on = (
+            pool.world > 1
+            and bool(getattr(get_config(), "stream_merge", True))
+            and not self._ipc_mode
+        )
+
+    def _discover_components(self) -> None:
+        """Find foldable pair_reduce components (called from run() once
+        task_deps is built)."""
+        if not self._stream_merge_on:
+            return
+        from lzy_amd.channels.treeplan import find_components
+
+        for ci, member_order in enumerate(find_components(self.calls)):
+            member_set = set(member_order)
+            entry_members = {
+                t for t in member_order
+                if not any(d in member_set for d in self.task_deps.get(t, ()))
+            }
+            self.components[ci] = {
+                "order": member_order,
+                "members": member_set,
+                "entry_members": entry_members,
+                "ready": set(),
+                "broken": False,
+                "launched": False,
+            }
+            for t in member_order:
+                self.stream_components[t] = ci
 
     # -- metadata helpers ---------------------------------------------------
 
@@ -901,6 +986,7 @@ class _DriverScheduler:
             self.task_deps[c.id] = deps
         self.entry_producer = producer
         dag.seal()
+        self._discover_components()
         METRICS.observe("lzy_graph_build", time.perf_counter() - t0)
 
         for tid in dag.take_ready():
@@ -949,6 +1035,10 @@ class _DriverScheduler:
         self.stopping = reason
         pool = self.pool
         self.deferred_gangs = []
+        for cid in self.deferred_plans:
+            for m in self.components[cid]["order"]:
+                self.journal.record(m, "cancelled", reason)
+        self.deferred_plans = []
         for tid in self.calls:
             if tid not in self.dispatched:
                 self.journal.record(tid, "cancelled", reason)
@@ -1061,6 +1151,8 @@ class _DriverScheduler:
                 self.inflight -= 1
                 self.task_dispatch_ts.pop(tid, None)
                 self.chained_waits.pop(tid, None)
+            elif ev in ("plan_done", "plan_failed"):
+                self._on_plan_event(rank, msg, dag, failed_tasks)
             elif ev == "log_chunk":
                 self._print_log_chunk(msg)
             elif ev == "task_cancelled":
@@ -1096,6 +1188,12 @@ class _DriverScheduler:
         if self.stopping is not None:
             self.journal.record(task_id, "cancelled", self.stopping)
             return
+        cid = self.stream_components.get(task_id)
+        if cid is not None and chain_rank is None:
+            comp = self.components[cid]
+            if not comp["broken"]:
+                self._plan_member_ready(task_id, comp, cid)
+                return
         call = self.calls[task_id]
         gpu_count = call.env.provisioning.effective_gpu_count
         pool = self.pool
@@ -1379,6 +1477,171 @@ class _DriverScheduler:
             # tasks were already failed by the worker_lost sweep
             pass
 
+    # -- streamed merge-tree plans ------------------------------------------
+
+    def _plan_member_ready(self, tid: str, comp: dict, cid: int) -> None:
+        """A component member became DAG-ready.  Members are held back
+        until every entry member is ready (interior members only become
+        ready through the plan itself), then the whole component launches
+        as one streamed plan — or breaks apart to op-by-op execution if
+        the leaves don't verify (shape/dtype mismatch etc.)."""
+        self.dispatched.add(tid)
+        comp["ready"].add(tid)
+        if not comp["entry_members"] <= comp["ready"] or comp["launched"]:
+            return
+        if self.active_plan is not None:
+            # one plan active pool-wide: per-(src,dst) p2p issue order on
+            # pg_stream must stay sequential
+            if cid not in self.deferred_plans:
+                self.deferred_plans.append(cid)
+            return
+        self._launch_plan(cid, comp)
+
+    def _break_component(self, comp: dict) -> None:
+        comp["broken"] = True
+        ready = sorted(comp["ready"])
+        comp["ready"] = set()
+        for m in ready:
+            self.dispatched.discard(m)
+        for m in ready:
+            if m not in self.dispatched:
+                self._dispatch(m)
+        self._try_chain()
+
+    def _launch_plan(self, cid: int, comp: dict) -> None:
+        from lzy_amd.channels.treeplan import build_plan
+
+        pool = self.pool
+
+        def meta_of(eid: str):
+            m = self.meta.get(eid)
+            if m is None:
+                try:
+                    m = self._meta_for_driver_entry(eid)
+                except KeyError:
+                    return None
+            # only CONFIRMED owners may stream (cf. transfer sourcing)
+            confirmed = {
+                o for o in m.owners
+                if (o, eid) not in self.transferred_now and o in self.outstanding
+            }
+            if not confirmed:
+                return None
+            import copy as _copy
+
+            m2 = _copy.copy(m)
+            m2.owners = confirmed
+            return m2
+
+        agent_tr = getattr(pool.agent, "transport", None)
+        plan = build_plan(
+            f"plan-{cid}-{pool.next_seq()}",
+            comp["order"], self.calls, meta_of,
+            chunk_bytes=self._stream_chunk,
+            cuda_p2p=bool(agent_tr is not None and agent_tr._cuda_p2p),
+        )
+        if plan is None:
+            _LOG.info("stream component %d not foldable — op-by-op", cid)
+            self._break_component(comp)
+            return
+        comp["launched"] = True
+        steps_by_rank = plan.pop("steps_by_rank")
+        t0 = time.perf_counter()
+        self.active_plan = {
+            "cid": cid,
+            "plan_id": plan["plan_id"],
+            "pending": set(plan["participants"]),
+            "node_rank": dict(plan["node_rank"]),
+            "outputs": [],
+            "failed": [],
+            "t0": t0,
+        }
+        from lzy_amd.config import get_config
+
+        wait_s = float(getattr(get_config(), "settle_wait_s", 120.0))
+        for r in plan["participants"]:
+            pool.driver_ctrl.send(r, {
+                "cmd": "stream_plan", "plan": plan,
+                "steps": steps_by_rank[r], "wait_timeout": wait_s,
+            })
+            if r in self.outstanding:
+                self.outstanding[r] += 1
+        self.inflight += 1
+        for m in comp["order"]:
+            self.journal.record(m, "scheduled", self.calls[m].callable_name)
+        # account the cross-rank edges as transfers (bench/SCALE telemetry)
+        n_edges = sum(
+            1 for sts in steps_by_rank.values() for st in sts
+            if st["op"] == "leaf_send"
+        ) + sum(
+            len(st.get("send_to", ())) for sts in steps_by_rank.values()
+            for st in sts
+        )
+        elem = torch.empty(0, dtype=getattr(torch, plan["dtype"])).element_size()
+        METRICS.inc("lzy_stream_plans")
+        METRICS.inc("lzy_transfers", n_edges)
+        METRICS.inc("lzy_transfer_bytes", n_edges * plan["numel"] * elem)
+        _LOG.info(
+            "stream plan %s: %d nodes on ranks %s, %d cross edges",
+            plan["plan_id"], len(comp["order"]), plan["participants"], n_edges,
+        )
+
+    def _on_plan_event(self, rank: int, msg: dict, dag,
+                       failed_tasks: Set[str]) -> None:
+        ap = self.active_plan
+        if ap is None or msg.get("plan_id") != ap["plan_id"]:
+            return  # stale
+        ev = msg["ev"]
+        ap["pending"].discard(rank)
+        if rank in self.outstanding:
+            self.outstanding[rank] -= 1
+        if ev == "plan_done":
+            ap["outputs"].extend(msg.get("outputs", ()))
+        else:
+            ap["failed"].append((rank, msg.get("error", "plan failed")))
+        if ap["pending"]:
+            return
+        # all participants reported: resolve the whole component
+        self.active_plan = None
+        self.inflight -= 1
+        comp = self.components[ap["cid"]]
+        if not ap["failed"]:
+            METRICS.observe("lzy_stream_plan_s", time.perf_counter() - ap["t0"])
+            by_task = {w["task"]: w for w in ap["outputs"]}
+            for m in comp["order"]:
+                wire = by_task.get(m)
+                if wire is not None:
+                    meta = EntryMeta.from_wire(wire)
+                    # the node's compute rank (from the plan builder)
+                    # owns the materialized accumulator
+                    meta.owners = {ap["node_rank"].get(m, 0)}
+                    self.meta[meta.entry_id] = meta
+                self.journal.record(m, "done")
+                self.dag_completed.add(m)
+                for nxt in dag.complete(m):
+                    if nxt not in self.dispatched:
+                        self._dispatch(nxt)
+            self._try_chain()
+        else:
+            rank0, err = ap["failed"][0]
+            for m in comp["order"]:
+                self.errors.append(LzyExecutionError(
+                    f"streamed merge plan failed on rank {rank0}: {err}",
+                    task_id=m,
+                ))
+                self.journal.record(m, "failed", err)
+                if m not in failed_tasks:
+                    failed_tasks.add(m)
+                    for ct in dag.fail(m):
+                        if ct not in comp["members"] and ct not in self.dispatched:
+                            self.journal.record(ct, "cancelled")
+        # a deferred plan may launch now
+        while self.deferred_plans and self.active_plan is None:
+            nxt_cid = self.deferred_plans.pop(0)
+            nxt_comp = self.components[nxt_cid]
+            if not nxt_comp["launched"] and not nxt_comp["broken"]:
+                self._launch_plan(nxt_cid, nxt_comp)
+
     # -- worker death --------------------------------------------------------
 
     def _on_worker_lost(self, rank: int, dag, failed_tasks: Set[str]) -> None:
@@ -1394,6 +1657,27 @@ class _DriverScheduler:
         whose only input copy died with the rank and has no durable blob.
         """
         self.outstanding.pop(rank, None)
+        # an active streamed plan touching the dead rank cannot complete:
+        # fail its members now (survivors' recv timeouts surface later as
+        # stale plan_failed events and are dropped)
+        ap = self.active_plan
+        if ap is not None and (
+            rank in ap["pending"] or rank in ap["node_rank"].values()
+        ):
+            self.active_plan = None
+            self.inflight -= 1
+            comp = self.components[ap["cid"]]
+            for m in comp["order"]:
+                self.errors.append(LzyExecutionError(
+                    f"worker rank {rank} died during streamed merge plan",
+                    task_id=m,
+                ))
+                self.journal.record(m, "failed", f"worker {rank} lost")
+                if m not in failed_tasks:
+                    failed_tasks.add(m)
+                    for ct in dag.fail(m):
+                        if ct not in comp["members"] and ct not in self.dispatched:
+                            self.journal.record(ct, "cancelled")
         for m in self.meta.values():
             if rank in m.owners:
                 m.owners.discard(rank)
@@ -1516,6 +1800,9 @@ class _DriverScheduler:
                 continue
             if call.env.provisioning.effective_gpu_count > 1:
                 continue  # gangs stay on the completion-driven path
+            cid = self.stream_components.get(tid)
+            if cid is not None and not self.components[cid]["broken"]:
+                continue  # folded into a streamed plan, never chained
             deps = self.task_deps.get(tid, [])
             if not deps or any(d not in self.dispatched for d in deps):
                 continue

@@ -109,6 +109,16 @@ def test_worker_death_detected_without_retries(tmp_path):
     assert "DEATH-RECOVERED" not in out0
 
 
+def test_streamed_merge_tree_world4(tmp_path):
+    """pair_reduce trees fold into one chunk-streamed plan; results match
+    the op-by-op reference; interior outputs stay readable; mismatched
+    components break apart gracefully."""
+    r = _run_distributed("tests/pool_script_streammerge.py", 4, tmp_path,
+                         timeout=300)
+    assert r.returncode == 0, r.stdout[-3000:] + r.stderr[-3000:]
+    assert "STREAMMERGE-OK" in r.stdout
+
+
 def test_live_remote_log_tail(tmp_path):
     """A remote rank's print appears on the driver console BEFORE the op
     finishes (live ReadStdSlots-style streaming, not completion-time)."""
