@@ -215,6 +215,24 @@ def run_dag(lzy: Lzy, width: int, step_idx: int) -> float:
     return out
 
 
+def _preflight(pool, n_gpus: int) -> None:
+    """Fail fast with a rank-tagged diagnostic instead of a silent hang:
+    every rank exercises every communicator (default, pg_data, pg_stream;
+    CPU and CUDA) once, so a broken RCCL rendezvous surfaces here, not
+    mid-benchmark.  Workers run it via the pool's preflight command."""
+    try:
+        pool.preflight()
+    except BaseException as e:
+        print(
+            f"[bench rank {pool.rank}] PREFLIGHT FAILED "
+            f"({type(e).__name__}: {e}) world={pool.world} "
+            f"device={pool.device} "
+            f"cuda_p2p={pool.agent.transport._cuda_p2p}",
+            file=sys.stderr, flush=True,
+        )
+        raise
+
+
 def main() -> None:
     parser = argparse.ArgumentParser()
     parser.add_argument("--gpus", type=int, default=1)
@@ -233,30 +251,80 @@ def main() -> None:
     from lzy_amd.runtime.pool import GpuPool, GpuPoolRuntime
 
     pool = GpuPool.get()  # workers serve here and never return
+    _preflight(pool, n_gpus)
     runtime = GpuPoolRuntime()
     lzy = Lzy(runtime=runtime)
 
+    t_w0 = time.perf_counter()
     for i in range(args.warmup):
         run_dag(lzy, width, i)
+    t_dag_est = (time.perf_counter() - t_w0) / max(1, args.warmup)
 
-    METRICS.reset()
-    ts_a = pool.sync_all()
-    t0 = time.perf_counter()
-    for i in range(args.steps):
-        run_dag(lzy, width, args.warmup + i)
-    ts_b = pool.sync_all()
-    t1 = time.perf_counter()
+    # the timed region must be long enough for utilization samplers and
+    # rocprof attribution to register (>= ~2 s): each *step* executes the
+    # DAG `repeats` times so K driver-chosen steps still span >= ~2.4 s.
+    # The reported metric stays the PER-DAG makespan; repeats is declared
+    # in config (honest accounting, same work per DAG).
+    repeats = max(1, min(50, int(0.12 / max(1e-4, t_dag_est))))
+
+    def step(i: int) -> None:
+        for r in range(repeats):
+            run_dag(lzy, width, args.warmup + i * repeats + r)
+
+    try:
+        METRICS.reset()
+        ts_a = pool.sync_all()
+        t0 = time.perf_counter()
+        for i in range(args.steps):
+            step(i)
+        ts_b = pool.sync_all()
+        t1 = time.perf_counter()
+    except BaseException as e:
+        import traceback
+
+        print(
+            f"[bench rank {pool.rank}] STEP FAILED: {type(e).__name__}: {e}\n"
+            + traceback.format_exc(),
+            file=sys.stderr, flush=True,
+        )
+        raise
 
     # per-rank elapsed between the two barriers, max over ranks
     elapsed_by_rank = {r: ts_b[r] - ts_a[r] for r in ts_b if r in ts_a}
     elapsed = max(max(elapsed_by_rank.values(), default=t1 - t0), t1 - t0)
 
-    makespan_s = elapsed / args.steps
+    n_dags = args.steps * repeats
+    makespan_s = elapsed / n_dags          # the metric: per-DAG makespan
     dispatch = METRICS.timing_stats("lzy_dispatch")
     sched_overhead_ms = 1000.0 * dispatch.get("mean", 0.0)
 
     shard_mb = (SHARD_MB or (1024 if torch.cuda.is_available() else 8))
     n_ops = 5 * width + (width - 1) + 2
+
+    transfers = int(METRICS.counter_value("lzy_transfers")
+                    + METRICS.counter_value("lzy_transfers_ipc"))
+    if width > 1 and transfers == 0:
+        raise RuntimeError(
+            f"N={width} but zero cross-rank transfers were recorded — "
+            "the data plane did not engage; refusing to report a number "
+            "that does not measure it"
+        )
+
+    # per-stage timing breakdown (rocprof/gpu_busy attribution aid)
+    stage_ms = {}
+    for name in sorted(getattr(METRICS, "_timings", {})):
+        if name.startswith("lzy_op::"):
+            st = METRICS.timing_stats(name)
+            stage_ms[name[len("lzy_op::"):]] = round(st["mean"] * 1e3, 3)
+
+    # effective per-edge bandwidth estimate from the streamed merge plans
+    # (plan time ~= one pipelined shard transfer)
+    plan_stats = METRICS.timing_stats("lzy_stream_plan_s")
+    xgmi_gbps_est = None
+    if plan_stats.get("count", 0) and plan_stats.get("mean", 0.0) > 0:
+        xgmi_gbps_est = round(
+            (shard_mb * (1 << 20)) / plan_stats["mean"] / 1e9, 2
+        )
 
     if os.environ.get("LZY_BENCH_PROFILE"):
         import sys as _sys
@@ -277,7 +345,9 @@ def main() -> None:
         "n_gpus": n_gpus,
         "steps": args.steps,
         "warmup": args.warmup,
-        "ms_per_step": round(1000.0 * makespan_s, 3),
+        # ms per driver step (repeats DAGs each) — steps * ms_per_step
+        # reproduces the timed-region wall clock for the driver's check
+        "ms_per_step": round(1000.0 * elapsed / args.steps, 3),
         "higher_is_better": False,
         "scaling": "weak",
         "vs_baseline": None,
@@ -291,15 +361,25 @@ def main() -> None:
             "parallelism": f"dag-fanout{width}",
             "shard_mb": shard_mb,
             "ops_per_dag": n_ops,
+            # one driver step = this many full DAG executions, so the
+            # timed region spans >= ~2 s for utilization sampling; the
+            # reported value is still the per-DAG makespan
+            "dag_repeats_per_step": repeats,
+            "dag_ms": round(1000.0 * makespan_s, 3),
+            "timed_region_s": round(elapsed, 3),
             "dispatch_p99_ms": round(1000.0 * dispatch.get("p99", 0.0), 4),
-            "transfers": int(METRICS.counter_value("lzy_transfers")
-                             + METRICS.counter_value("lzy_transfers_ipc")),
+            "transfers": transfers,
             "transfer_gb": round(
                 METRICS.counter_value("lzy_transfer_bytes") / 1e9, 3
             ),
             "chain_dispatches": int(
                 METRICS.counter_value("lzy_chain_dispatches")
             ),
+            "stream_plans": int(METRICS.counter_value("lzy_stream_plans")),
+            "stream_plan_ms": round(plan_stats.get("mean", 0.0) * 1e3, 3)
+            if plan_stats.get("count") else None,
+            "xgmi_gbps_est": xgmi_gbps_est,
+            "stage_ms": stage_ms,
         },
     }
     print(json.dumps(result), flush=True)

@@ -172,7 +172,14 @@ class WorkerAgent:
             for item in msg["items"]:
                 eid = item["entry"]
                 if not self.store.has(eid):  # defense: settle may be landing
-                    self.store.wait_present(eid, timeout=30.0)
+                    # config-derived: a multi-GB fan-in over one xGMI link
+                    # can legitimately outlast a fixed small bound
+                    from lzy_amd.config import get_config
+
+                    self.store.wait_present(
+                        eid,
+                        timeout=float(getattr(get_config(), "settle_wait_s", 120.0)),
+                    )
                 value = self.store.get(eid)
                 ops, keep = self.transport.send_ops(
                     value, self.store.pickled.get(eid), item["dst"]
@@ -233,6 +240,8 @@ class WorkerAgent:
             self.ctrl.send_event({"ev": "ack", "tag": tag, "rank": self.rank})
         elif cmd == "barrier":
             self._exec_q.put(msg)
+        elif cmd == "preflight":
+            self._exec_q.put(msg)
         elif cmd == "load_serializers":
             self.serializers.load_user_serializers(msg["payload"])
             self.ctrl.send_event({"ev": "ack", "tag": msg["tag"], "rank": self.rank})
@@ -252,6 +261,31 @@ class WorkerAgent:
         elif cmd == "shutdown":
             self._shutdown = True
             self._exec_q.put({"cmd": "_stop"})
+
+    def run_preflight(self) -> None:
+        """Exercise every communicator once (default, pg_data, pg_stream;
+        CPU and, on the RCCL path, CUDA) so a broken rendezvous fails
+        fast with a rank-tagged error instead of hanging mid-benchmark.
+        All ranks must run this concurrently (collectives)."""
+        groups = [("default", None), ("pg_data", self.transport._pg),
+                  ("pg_stream", self.pg_stream)]
+        for name, group in groups:
+            t = torch.ones(1)
+            dist.all_reduce(t, group=group)
+            if int(t.item()) != self.world:
+                raise RuntimeError(
+                    f"preflight {name}: CPU all_reduce gave {t.item()}, "
+                    f"want {self.world}"
+                )
+            if self.device is not None and self.transport._cuda_p2p:
+                d = torch.ones(1, device=self.device)
+                dist.all_reduce(d, group=group)
+                torch.cuda.synchronize(self.device)
+                if int(d.item()) != self.world:
+                    raise RuntimeError(
+                        f"preflight {name}: CUDA all_reduce gave {d.item()}, "
+                        f"want {self.world}"
+                    )
 
     def _prune_outbox(self) -> None:
         still = []
@@ -274,6 +308,14 @@ class WorkerAgent:
                 with self._futs_lock:
                     self._futs.add(fut)
                 fut.add_done_callback(self._fut_done)
+            elif cmd == "preflight":
+                try:
+                    self.run_preflight()
+                    self.ctrl.send_event(
+                        {"ev": "ack", "tag": msg["tag"], "rank": self.rank}
+                    )
+                except BaseException as e:  # noqa: BLE001
+                    self._report_error(msg, e)
             elif cmd == "barrier":
                 # a barrier orders after every previously submitted task
                 self._drain_tasks()
@@ -609,7 +651,11 @@ class GpuPool:
             self.events.put((rank, msg))
 
     def wait_acks(self, kind: str, tag: str, ranks: Sequence[int],
-                  timeout: float = 300.0) -> Dict[int, Any]:
+                  timeout: Optional[float] = None) -> Dict[int, Any]:
+        if timeout is None:
+            from lzy_amd.config import get_config
+
+            timeout = float(getattr(get_config(), "ack_wait_s", 300.0))
         key = f"{kind}:{tag}"
         want = set(ranks)
         with self._ack_cv:
@@ -637,6 +683,17 @@ class GpuPool:
             self.wait_acks("ack", tag, range(self.world))
             self._group_tags.add(tag)
         return tag
+
+    def preflight(self) -> None:
+        """Driver-side: run the communicator preflight on every rank
+        (fail-fast RCCL/gloo diagnostics — all ranks join the
+        collectives; a failure surfaces as a rank-tagged TimeoutError or
+        agent error instead of a silent mid-benchmark hang)."""
+        if self.world <= 1 or not self.is_driver:
+            return
+        tag = f"pf{self.next_seq()}"
+        self.driver_ctrl.broadcast({"cmd": "preflight", "tag": tag})
+        self.wait_acks("ack", tag, range(self.world), timeout=240.0)
 
     def sync_all(self) -> Dict[int, float]:
         """Barrier across all ranks (through exec queues, so it orders after

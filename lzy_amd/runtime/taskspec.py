@@ -99,9 +99,14 @@ class WorkerStore:
         self.spill.track(entry_id, value)
         self.spill.maybe_spill(self.values)
 
-    def wait_present(self, entry_id: str, timeout: float = 120.0) -> bool:
+    def wait_present(self, entry_id: str, timeout: Optional[float] = None) -> bool:
         """Block until the entry lands in the store (another task's settle
-        may be completing the transfer concurrently)."""
+        may be completing the transfer concurrently).  Default timeout is
+        config-derived (settle_wait_s)."""
+        if timeout is None:
+            from lzy_amd.config import get_config
+
+            timeout = float(getattr(get_config(), "settle_wait_s", 120.0))
         with self._cond:
             return self._cond.wait_for(
                 lambda: entry_id in self.values, timeout=timeout
