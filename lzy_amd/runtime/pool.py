@@ -978,10 +978,14 @@ class _DriverScheduler:
             and agent_tr is not None
             and not agent_tr._cuda_p2p
         )
+        # plans stay on under AUTO-ipc (oversubscribed harness: the plan
+        # executor host-stages CUDA chunks through gloo), so the 1-GPU
+        # rehearsal exercises the same schedule the 8-GPU RCCL run uses;
+        # an EXPLICIT ipc transport choice disables them
         self._stream_merge_on = (
             pool.world > 1
             and bool(getattr(get_config(), "stream_merge", True))
-            and not self._ipc_mode
+            and not ipc_enabled()
         )
 
     def _discover_components(self) -> None:
