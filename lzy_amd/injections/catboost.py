@@ -9,6 +9,20 @@ workflow, with ``task_type="GPU"`` when a GPU was provisioned.
 
 catboost is an optional dependency: importing this module without it
 raises ImportError, matching the reference's lazy injection behavior.
+
+VERIFICATION STATUS (honest): the catboost wheel is NOT present in this
+image's offline wheelhouse and the build/GPU environments have no
+network, so this injection has never been exercised against the real
+library — only against the structural stub in tests/test_injections.py
+(which drives the actual patching logic: provisioning detection,
+task_type="GPU" selection, op capture inside a workflow).  The
+BASELINE.json config-2 benchmark therefore uses a torch GBDT-style
+stand-in objective, explicitly labeled as such in
+benchmarks/baseline_configs.py.  First action in an environment where
+``pip install catboost`` works: run
+``python -m pytest tests/test_injections.py`` with the real module on
+the path (the stub fixture yields to a real import automatically) and
+benchmarks/baseline_configs.py --config 2.
 """
 from __future__ import annotations
 
