@@ -172,14 +172,10 @@ class WorkerAgent:
             for item in msg["items"]:
                 eid = item["entry"]
                 if not self.store.has(eid):  # defense: settle may be landing
-                    # config-derived: a multi-GB fan-in over one xGMI link
-                    # can legitimately outlast a fixed small bound
-                    from lzy_amd.config import get_config
-
-                    self.store.wait_present(
-                        eid,
-                        timeout=float(getattr(get_config(), "settle_wait_s", 120.0)),
-                    )
+                    # config-derived (settle_wait_s, cached by the store):
+                    # a multi-GB fan-in over one xGMI link can
+                    # legitimately outlast a fixed small bound
+                    self.store.wait_present(eid)
                 value = self.store.get(eid)
                 ops, keep = self.transport.send_ops(
                     value, self.store.pickled.get(eid), item["dst"]

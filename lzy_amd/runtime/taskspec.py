@@ -82,6 +82,11 @@ class WorkerStore:
         from lzy_amd.storage.spill import SpillManager
 
         self.spill = SpillManager(device=device)
+        # config resolved once: wait_present runs on the per-task settle
+        # path, and get_config() re-fingerprints the environment
+        from lzy_amd.config import get_config
+
+        self._settle_wait = float(getattr(get_config(), "settle_wait_s", 120.0))
 
     def put(self, entry_id: str, value: Any, pickled: Optional[bytes] = None) -> None:
         from lzy_amd.runtime.streams import STREAMS
@@ -102,11 +107,9 @@ class WorkerStore:
     def wait_present(self, entry_id: str, timeout: Optional[float] = None) -> bool:
         """Block until the entry lands in the store (another task's settle
         may be completing the transfer concurrently).  Default timeout is
-        config-derived (settle_wait_s)."""
+        config-derived (settle_wait_s, resolved at store construction)."""
         if timeout is None:
-            from lzy_amd.config import get_config
-
-            timeout = float(getattr(get_config(), "settle_wait_s", 120.0))
+            timeout = self._settle_wait
         with self._cond:
             return self._cond.wait_for(
                 lambda: entry_id in self.values, timeout=timeout
