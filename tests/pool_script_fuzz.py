@@ -57,6 +57,14 @@ def cached_square(x: float) -> float:
     return x * x
 
 
+@op(pair_reduce=(0.5, 0.5))
+def tmerge(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    # pair_reduce edges form random trees/chains the scheduler may fold
+    # into streamed plans (or break apart on shape mismatch) — both
+    # paths must match this body's semantics exactly
+    return (a + b) * 0.5
+
+
 @op(gpu_count=2)
 def gang_sum2(x: float) -> float:
     import torch.distributed as dist
@@ -98,10 +106,20 @@ def build_and_run(lzy, seed: int):
             elif kind < 0.65 and int(os.environ.get("WORLD_SIZE", "1")) >= 2:
                 p, e = rng.choice(scalars)
                 scalars.append((gang_sum2(p), e * 2.0))
-            elif kind < 0.8 and len(tensors) >= 2:
+            elif kind < 0.73 and len(tensors) >= 2:
                 (pa, ea), (pb, eb) = rng.sample(tensors, 2)
                 m = min(ea.numel(), eb.numel())
                 tensors.append((tcombine(pa, pb), ea[:m] + eb[:m]))
+            elif kind < 0.8 and len(tensors) >= 2:
+                (pa, ea), (pb, eb) = rng.sample(tensors, 2)
+                if ea.numel() == eb.numel():
+                    # foldable pair-reduce edge (streamed plan candidate)
+                    tensors.append((tmerge(pa, pb), (ea + eb) * 0.5))
+                else:
+                    # mismatched shapes: the fold must break and still
+                    # execute the op body correctly
+                    m = min(ea.numel(), eb.numel())
+                    tensors.append((tcombine(pa, pb), ea[:m] + eb[:m]))
             else:
                 p, e = rng.choice(tensors)
                 scalars.append((tsum(p), float(e.sum())))
