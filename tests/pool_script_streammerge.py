@@ -24,7 +24,10 @@ N = 1 << 20  # 4 MiB f32 shards -> several 1 MiB-min chunks
 @op
 def make_shard(i: int) -> torch.Tensor:
     g = torch.Generator().manual_seed(1000 + i)
-    return torch.randn(N, generator=g)
+    t = torch.randn(N, generator=g)
+    if torch.cuda.is_available():
+        t = t.cuda()  # plan executor host-stages chunks on 1-GPU boxes
+    return t
 
 
 @op
