@@ -126,6 +126,12 @@ class WorkerAgent:
         self._task_pool = ThreadPoolExecutor(
             max_workers=n_exec, thread_name_prefix=f"lzy-task-r{rank}"
         )
+        # streamed plans are communication-bound and sit on the makespan
+        # critical path: a dedicated thread keeps them from queueing
+        # behind compute tasks (plans are serialized pool-wide anyway)
+        self._plan_pool = ThreadPoolExecutor(
+            max_workers=1, thread_name_prefix=f"lzy-plan-r{rank}"
+        )
         self._futs_lock = threading.Lock()
         self._futs: Set[Any] = set()
         self._exec_thread = threading.Thread(
@@ -298,9 +304,14 @@ class WorkerAgent:
             cmd = msg["cmd"]
             if cmd == "_stop":
                 self._task_pool.shutdown(wait=False)
+                self._plan_pool.shutdown(wait=False)
                 return
             if cmd in ("task", "settle", "stream_plan"):
-                fut = self._task_pool.submit(self._run_guarded, msg)
+                pool = (
+                    self._plan_pool if cmd == "stream_plan"
+                    else self._task_pool
+                )
+                fut = pool.submit(self._run_guarded, msg)
                 with self._futs_lock:
                     self._futs.add(fut)
                 fut.add_done_callback(self._fut_done)
