@@ -74,6 +74,11 @@ def _try_load() -> Optional[ctypes.CDLL]:
         ctypes.c_void_p, ctypes.c_int, ctypes.c_int64, ctypes.c_int,
         ctypes.c_void_p, ctypes.c_void_p,
     ]
+    lib.lz_stats_variant.restype = ctypes.c_int
+    lib.lz_stats_variant.argtypes = [
+        ctypes.c_void_p, ctypes.c_int, ctypes.c_int64, ctypes.c_int,
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int,
+    ]
     lib.lz_normalize_apply.restype = ctypes.c_int
     lib.lz_normalize_apply.argtypes = [
         ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p, ctypes.c_int,
@@ -227,6 +232,28 @@ def stats(t, use_abs: bool = False):
             1 if use_abs else 0,
             ctypes.c_void_p(out.data_ptr()),
             ctypes.c_void_p(_current_stream_ptr()),
+        )
+    )
+    return out
+
+
+def stats_variant(t, variant: int, use_abs: bool = False):
+    """Sweep-only: run a (V, U) load-shape variant of the stats kernel
+    (benchmarks/kernel_sweep.py picks the default)."""
+    import torch
+
+    lib = _require_native()
+    flat = t.detach().contiguous()
+    out = torch.empty(2, dtype=torch.float64, device=t.device)
+    _check(
+        lib.lz_stats_variant(
+            ctypes.c_void_p(flat.data_ptr()),
+            _dtype_code(flat.dtype),
+            flat.numel(),
+            1 if use_abs else 0,
+            ctypes.c_void_p(out.data_ptr()),
+            ctypes.c_void_p(_current_stream_ptr()),
+            int(variant),
         )
     )
     return out

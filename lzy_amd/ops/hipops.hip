@@ -356,13 +356,13 @@ __device__ __forceinline__ double lz_wave_reduce_f64(double v) {
 // flight) with separate accumulator pairs — a single accumulate chain
 // leaves the reduction HBM-latency-bound (measured 2.8 TB/s; the
 // streaming kernels hit 5.4+).
-template <typename T, bool ABS>
+template <typename T, bool ABS,
+          int V = (sizeof(T) == 2) ? 16 : 8, int U = 4>
 __global__ void stats_kernel(const T* __restrict__ src, int64_t n,
                              double* __restrict__ out) {
-    // 32 B per lane per vector load regardless of dtype width: 16-bit
-    // inputs at V=8 (16 B) leave the read stream request-starved
-    constexpr int V = (sizeof(T) == 2) ? 16 : 8;
-    constexpr int U = 4;
+    // Default 32 B per lane per vector load regardless of dtype width:
+    // 16-bit inputs at V=8 (16 B) leave the read stream request-starved.
+    // V/U are template-swept (lz_stats_variant + benchmarks/kernel_sweep)
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     int64_t vec_n = n / V;
     using SrcV = struct { T v[V]; };
@@ -533,6 +533,62 @@ extern "C" hipError_t lz_stats(const void* src, int dtype, int64_t n, int use_ab
         default: return hipErrorInvalidValue;
     }
 #undef LZ_STATS_CASE
+    return hipGetLastError();
+}
+
+// sweep-only entry: V/U variants of the stats reduction (the one
+// data-plane kernel measurably below the reduction ceiling — checksum
+// streams at ~4.2-4.6 TB/s, stats at ~3.2).  Variants are compiled for
+// the 16-bit path (bf16/f16); best found becomes the default above.
+extern "C" hipError_t lz_stats_variant(const void* src, int dtype, int64_t n,
+                                       int use_abs, double* out_device,
+                                       void* stream, int variant) {
+    hipStream_t s = (hipStream_t)stream;
+    hipError_t err = hipMemsetAsync(out_device, 0, 16, s);
+    if (err != hipSuccess) return err;
+    int blocks = lz_grid_for(n / 8);
+#define LZ_SV_LAUNCH(T, V, U)                                                   \
+    do {                                                                        \
+        if (use_abs)                                                            \
+            hipLaunchKernelGGL((stats_kernel<T, true, V, U>), dim3(blocks),     \
+                               dim3(LZ_BLOCK), 0, s, (const T*)src, n,          \
+                               out_device);                                     \
+        else                                                                    \
+            hipLaunchKernelGGL((stats_kernel<T, false, V, U>), dim3(blocks),    \
+                               dim3(LZ_BLOCK), 0, s, (const T*)src, n,          \
+                               out_device);                                     \
+    } while (0)
+#define LZ_SV_DTYPE(T)                                                          \
+    switch (variant) {                                                          \
+        case 0: LZ_SV_LAUNCH(T, 16, 4); break;                                  \
+        case 1: LZ_SV_LAUNCH(T, 16, 8); break;                                  \
+        case 2: LZ_SV_LAUNCH(T, 32, 2); break;                                  \
+        case 3: LZ_SV_LAUNCH(T, 32, 4); break;                                  \
+        case 4: LZ_SV_LAUNCH(T, 8, 8); break;                                   \
+        case 5: LZ_SV_LAUNCH(T, 16, 2); break;                                  \
+        case 6: LZ_SV_LAUNCH(T, 8, 4); break;                                   \
+        default: return hipErrorInvalidValue;                                   \
+    }
+    switch (dtype) {
+        case LZ_F16: { LZ_SV_DTYPE(__half) break; }
+        case LZ_BF16: { LZ_SV_DTYPE(__hip_bfloat16) break; }
+        case LZ_F32: {
+            switch (variant) {
+                case 0: LZ_SV_LAUNCH(float, 8, 4); break;
+                case 1: LZ_SV_LAUNCH(float, 8, 8); break;
+                case 2: LZ_SV_LAUNCH(float, 16, 2); break;
+                case 3: LZ_SV_LAUNCH(float, 16, 4); break;
+                case 4: LZ_SV_LAUNCH(float, 4, 8); break;
+                case 5: LZ_SV_LAUNCH(float, 8, 2); break;
+                case 6: LZ_SV_LAUNCH(float, 4, 4); break;
+                default: return hipErrorInvalidValue;
+            }
+            break;
+        }
+        default: return hipErrorInvalidValue;
+    }
+#undef LZ_SV_DTYPE
+#undef LZ_SV_LAUNCH
     return hipGetLastError();
 }
 

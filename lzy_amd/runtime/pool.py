@@ -950,10 +950,15 @@ class _DriverScheduler:
         self.chained_waits: Dict[str, Tuple[int, List[str]]] = {}
         from lzy_amd.config import get_config
 
-        self._chain_enabled = bool(getattr(get_config(), "chain_dispatch", True))
+        # ONE config resolve per batch: get_config() re-fingerprints the
+        # LZY_* environment, which costs ~70 us a call — 8+ calls per
+        # two-batch DAG showed up in the host profile
+        cfg = get_config()
+        self._settle_wait = float(getattr(cfg, "settle_wait_s", 120.0))
+        self._chain_enabled = bool(getattr(cfg, "chain_dispatch", True))
         # worker-death recovery: per-task re-dispatch budget (reference:
         # scheduler re-allocation after VM death + storage-peer failover)
-        self._max_retries = int(getattr(get_config(), "task_retries", 1))
+        self._max_retries = int(getattr(cfg, "task_retries", 1))
         self.retry_budget: Dict[str, int] = {}
         self.retrying: Set[str] = set()
         # mid-flight StopGraph: reason once set; done event lets an
@@ -970,17 +975,14 @@ class _DriverScheduler:
         self.components: Dict[int, dict] = {}
         self.active_plan: Optional[dict] = None
         self.deferred_plans: List[int] = []
-        self._stream_chunk = int(
-            getattr(get_config(), "stream_chunk_mb", 64)
-        ) << 20
-        from lzy_amd.channels.transport import ipc_enabled
-
+        self._stream_chunk = int(getattr(cfg, "stream_chunk_mb", 64)) << 20
         # explicit ipc mode, or automatic when ranks outnumber GPUs:
         # RCCL cannot build a comm then, and the host-staged fallback is
         # ~19x slower than hipIpc zero-copy for device tensors on one
         # node (profiles/bench_history.md)
+        explicit_ipc = cfg.channel_transport == "ipc"
         agent_tr = getattr(pool.agent, "transport", None)
-        self._ipc_mode = ipc_enabled() or (
+        self._ipc_mode = explicit_ipc or (
             torch.cuda.is_available()
             and agent_tr is not None
             and not agent_tr._cuda_p2p
@@ -991,8 +993,8 @@ class _DriverScheduler:
         # an EXPLICIT ipc transport choice disables them
         self._stream_merge_on = (
             pool.world > 1
-            and bool(getattr(get_config(), "stream_merge", True))
-            and not ipc_enabled()
+            and bool(getattr(cfg, "stream_merge", True))
+            and not explicit_ipc
         )
 
     def _discover_components(self) -> None:
@@ -1624,9 +1626,7 @@ class _DriverScheduler:
             "failed": [],
             "t0": t0,
         }
-        from lzy_amd.config import get_config
-
-        wait_s = float(getattr(get_config(), "settle_wait_s", 120.0))
+        wait_s = self._settle_wait
         for r in plan["participants"]:
             pool.driver_ctrl.send(r, {
                 "cmd": "stream_plan", "plan": plan,
