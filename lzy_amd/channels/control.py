@@ -44,6 +44,11 @@ class DriverControl:
         self._lock = threading.Lock()
         self._threads: List[threading.Thread] = []
         self._closed = False
+        # rank 0 lives in THIS process: once set, its commands bypass the
+        # loopback socket (no pickle, no serve-thread hop) — ~2 hops and
+        # two (de)serializations saved per task and per transfer command.
+        # Per-rank command order is preserved (inline = program order).
+        self.local_handler: Optional[Callable[[dict], None]] = None
 
     @property
     def address(self) -> str:
@@ -91,16 +96,24 @@ class DriverControl:
                 self._on_event(rank, {"ev": "worker_lost"})
 
     def send(self, rank: int, msg: dict) -> None:
+        if rank == 0 and self.local_handler is not None:
+            self.local_handler(msg)
+            return
         with self._lock:
             self._conns[rank].send(msg)
 
     def broadcast(self, msg: dict) -> None:
+        local = self.local_handler
         with self._lock:
             for rank, conn in self._conns.items():
+                if rank == 0 and local is not None:
+                    continue  # delivered inline below
                 try:
                     conn.send(msg)
                 except (OSError, BrokenPipeError):
                     _LOG.warning("broadcast to dead rank %d skipped", rank)
+        if local is not None:
+            local(msg)
 
     def close(self) -> None:
         self._closed = True
@@ -119,8 +132,12 @@ class WorkerControl:
     def __init__(self, rank: int, address: str):
         import lzy_amd
 
+        self._rank = rank
         self._conn = Client(address, family="AF_UNIX", authkey=_AUTHKEY)
         self._send_lock = threading.Lock()
+        # driver-process rank 0: events flow straight into the pool's
+        # event sink instead of through the loopback socket
+        self.local_sink: Optional[Callable[[int, dict], None]] = None
         self._conn.send({
             "rank": rank,
             "version": lzy_amd.__version__,
@@ -131,6 +148,10 @@ class WorkerControl:
         return self._conn.recv()
 
     def send_event(self, msg: dict) -> None:
+        sink = self.local_sink
+        if sink is not None:
+            sink(self._rank, msg)
+            return
         with self._send_lock:
             self._conn.send(msg)
 

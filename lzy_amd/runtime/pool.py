@@ -168,6 +168,16 @@ class WorkerAgent:
                                msg.get("cmd"))
                 self._report_error(msg, e)
 
+    def handle_local(self, msg: dict) -> None:
+        """Inline command path for the driver-process rank 0 (no socket
+        hop); same error reporting as the serve loop."""
+        try:
+            self._handle(msg)
+        except BaseException as e:  # noqa: BLE001 - must not kill the caller
+            _LOG.exception("agent r%d local-handle error on %s", self.rank,
+                           msg.get("cmd"))
+            self._report_error(msg, e)
+
     def _handle(self, msg: dict) -> None:
         cmd = msg["cmd"]
         if cmd == "task":
@@ -627,6 +637,12 @@ class GpuPool:
 
         if self.is_driver:
             accept_thread.join()
+            # short-circuit the loopback: rank 0 lives in this process,
+            # so its commands and events skip the AF_UNIX socket (two
+            # pickle hops saved per task/transfer/event).  The serve
+            # thread stays up but idles — the socket goes silent.
+            self.driver_ctrl.local_handler = self.agent.handle_local
+            self.agent.ctrl.local_sink = self._on_event
             serve_thread = threading.Thread(
                 target=self.agent.serve_forever, daemon=True, name="lzy-agent-serve"
             )
