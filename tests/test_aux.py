@@ -259,6 +259,21 @@ def test_status_endpoint(lzy, storage_root, tmp_path):
     )
     assert isinstance(wbs, list)
 
+    # per-task drill-down (the reference SPA's task table)
+    eid = next(
+        w["execution_id"] for w in wfs if w["execution_id"].startswith("statuswf")
+    )
+    tasks = json.loads(
+        urllib.request.urlopen(f"http://127.0.0.1:{port}/workflows/{eid}").read()
+    )
+    assert tasks and tasks[0]["name"].endswith("one")
+    assert tasks[0]["state"] == "done"
+
+    gpus = json.loads(
+        urllib.request.urlopen(f"http://127.0.0.1:{port}/gpus").read()
+    )
+    assert isinstance(gpus, list)  # [] on a CPU box; populated on GPU
+
 
 def test_status_dashboard_html(storage_root):
     import urllib.request
