@@ -52,7 +52,12 @@ class StepGraph:
         self.captured = True
 
     def run(self, *args: torch.Tensor) -> torch.Tensor:
-        if self.fallback_eager:
+        # CPU-tensor steps must run eagerly even when a GPU exists:
+        # capturing them yields an EMPTY graph (no stream ops), and its
+        # replay would keep returning the capture-time result forever
+        if self.fallback_eager or not all(
+            isinstance(a, torch.Tensor) and a.is_cuda for a in args
+        ):
             return self._fn(*args)
         try:
             if self._graph is None or self._signature(args) != self._sig:

@@ -53,10 +53,19 @@ def test_stepgraph_eager_fallback_cpu():
         return loss.detach()
 
     sg = StepGraph(step)
-    a = sg.run(torch.randn(4, 8))
-    b = sg.run(torch.randn(4, 8))
+    xa, xb = torch.randn(4, 8), torch.randn(4, 8)
+    a = sg.run(xa)
+    b = sg.run(xb)
     assert a.item() >= 0 and b.item() >= 0
-    assert sg.fallback_eager and not sg.captured
+    # CPU-tensor steps NEVER capture — on a GPU box a capture would
+    # yield an empty graph whose replay returns stale results
+    assert not sg.captured
+    # and the step really re-executes per call (not a stale static out):
+    # SGD moves the weights between calls, so the same input gives a
+    # different loss
+    c = sg.run(xa).item()
+    d = sg.run(xa).item()
+    assert c != d
 
 
 def test_ops_cpu_guards():
