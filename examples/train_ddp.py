@@ -25,7 +25,12 @@ def train_epoch(dim: int, batch: int, steps: int) -> float:
     from lzy_amd.runtime.context import op_context
 
     ctx = op_context()
-    dev = torch.device("cuda") if torch.cuda.is_available() else torch.device("cpu")
+    # one rank per GPU is the production shape; in oversubscribed
+    # harnesses (more ranks than GPUs) RCCL refuses two ranks on one
+    # device, so collectives fall back to CPU/gloo there
+    world = len(ctx.ranks) if ctx is not None else 1
+    use_cuda = torch.cuda.is_available() and torch.cuda.device_count() >= world
+    dev = torch.device("cuda") if use_cuda else torch.device("cpu")
     torch.manual_seed(1234 + (ctx.gang_rank if ctx else 0))
 
     model = torch.nn.Sequential(
