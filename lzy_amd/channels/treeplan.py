@@ -61,6 +61,10 @@ def find_components(calls: Dict[str, Any]) -> List[List[str]]:
         and len(c.entry_ids) == 1
         and len(c.input_entry_ids()) == 2
         and not c.kwarg_entry_ids
+        # cached ops keep the op-by-op path: the plan executor bypasses
+        # the result cache (both lookup and write), and silently ignoring
+        # @op(cache=True) would break the caching contract
+        and not getattr(c, "cache", False)
     }
     if not members:
         return []

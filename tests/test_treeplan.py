@@ -72,6 +72,18 @@ def test_find_components_tree_and_singletons():
     assert find_components(single) == []
 
 
+def test_find_components_skips_cached_ops():
+    """cache=True ops must run op-by-op — the plan bypasses the result
+    cache entirely, which would silently break the caching contract."""
+    calls, _ = make_tree(4)
+    some = next(iter(calls))
+    calls[some].cache = True
+    comps = find_components(calls)
+    # the remaining 2 merges still form a component only if connected;
+    # with one level-0 merge removed the root links the other two
+    assert all(some not in c for c in comps)
+
+
 def test_find_components_skips_undeclared_and_malformed():
     FakeCall._n = 0
     calls = {
