@@ -41,7 +41,9 @@ class LzyCall:
         lazy_arguments: bool = False,
         pair_reduce=None,
     ) -> None:
-        self.id = str(uuid.uuid4())
+        from lzy_amd.utils.ids import fast_uid
+
+        self.id = fast_uid()
         self.workflow = workflow
         self.signature = signature
         self.output_types = tuple(output_types)

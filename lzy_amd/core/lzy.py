@@ -10,6 +10,7 @@ from __future__ import annotations
 import datetime
 import inspect
 import os
+import sys
 from dataclasses import dataclass, field
 from functools import cached_property
 from typing import Any, Iterable, Optional, Sequence
@@ -143,7 +144,9 @@ class Lzy(WithEnvironmentMixin):
         interactive: bool = True,
         env: Optional[LzyEnvironment] = None,
     ) -> LzyWorkflow:
-        frame = inspect.stack()[1].frame
+        # sys._getframe(1) is the caller frame without inspect.stack()'s
+        # full walk + source loading (findsource per frame — milliseconds)
+        frame = sys._getframe(1)
         namespace = {**frame.f_globals, **frame.f_locals}
         wf_env = env or LzyEnvironment()
         wf_env = self.env.combine(

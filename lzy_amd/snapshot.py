@@ -122,7 +122,9 @@ class DefaultSnapshot:
     def create_entry(
         self, name: str, typ: Type, storage_uri: Optional[str] = None
     ) -> SnapshotEntry:
-        eid = str(uuid.uuid4())
+        from lzy_amd.utils.ids import fast_uid
+
+        eid = fast_uid()
         ser = self._serializers.find_serializer_by_type(typ if isinstance(typ, type) else object)
         fmt = ser.data_format() if ser else "pickle"
         uri = storage_uri or f"{self._prefix}/{eid}.{name}"
