@@ -77,9 +77,43 @@ def gang_sum2(x: float) -> float:
     return float(t.item())
 
 
-def build_and_run(lzy, seed: int):
+def build_and_run(lzy, seed: int, chaos_abort: bool = False):
+    """One random DAG; with ``chaos_abort`` a side thread aborts the
+    workflow after a random delay — the run must either complete with
+    CORRECT values or raise WorkflowAbortedError promptly, and the pool
+    must stay usable for the next seed (StopGraph under fuzz)."""
+    import threading
+
+    from lzy_amd.exceptions import WorkflowAbortedError
+
     rng = random.Random(seed)
-    with lzy.workflow(f"fuzz-{seed}", interactive=False):
+    aborter = None
+    try:
+        with lzy.workflow(f"fuzz-{seed}", interactive=False) as _wf:
+            if chaos_abort and rng.random() < 0.5:
+                delay = rng.uniform(0.0, 0.15)
+
+                def _abort(wf=_wf, d=delay):
+                    import time as _t
+
+                    _t.sleep(d)
+                    try:
+                        wf.abort("chaos")
+                    except WorkflowAbortedError:
+                        pass
+
+                aborter = threading.Thread(target=_abort, daemon=True)
+                aborter.start()
+            _build_and_check(rng, seed)
+    except WorkflowAbortedError:
+        pass  # clean abort is a valid outcome
+    finally:
+        if aborter is not None:
+            aborter.join(timeout=60)
+
+
+def _build_and_check(rng, seed):
+    if True:
         scalars = []  # (proxy, expected)
         tensors = []  # (proxy, expected torch tensor on cpu)
         for i in range(rng.randint(2, 4)):
@@ -137,8 +171,9 @@ def main() -> None:
     lzy = Lzy(runtime=GpuPoolRuntime())
     base = int(os.environ.get("FUZZ_BASE_SEED", "1000"))
     rounds = int(os.environ.get("FUZZ_ROUNDS", "12"))
+    chaos = os.environ.get("FUZZ_CHAOS_ABORT", "") not in ("", "0")
     for k in range(rounds):
-        build_and_run(lzy, base + k)
+        build_and_run(lzy, base + k, chaos_abort=chaos)
     print("FUZZ-OK", flush=True)
 
 

@@ -224,3 +224,17 @@ def test_pool_fuzz_random_dags(world, tmp_path):
     )
     assert r.returncode == 0, r.stdout[-3000:] + r.stderr[-3000:]
     assert "FUZZ-OK" in r.stdout
+
+
+def test_pool_fuzz_chaos_abort(tmp_path):
+    """Random mid-flight aborts over the DAG fuzz: every run either
+    completes with correct sinks or raises WorkflowAbortedError, and the
+    pool stays usable across seeds (StopGraph under fuzz)."""
+    r = _run_distributed(
+        "tests/pool_script_fuzz.py", 2, tmp_path,
+        extra_env={"FUZZ_BASE_SEED": "4100", "FUZZ_ROUNDS": "12",
+                   "FUZZ_CHAOS_ABORT": "1"},
+        timeout=300,
+    )
+    assert r.returncode == 0, r.stdout[-3000:] + r.stderr[-3000:]
+    assert "FUZZ-OK" in r.stdout
