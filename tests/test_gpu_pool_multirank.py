@@ -47,3 +47,32 @@ def test_pool_two_ranks_one_gpu(tmp_path, transport):
         print("STDERR:", res.stderr[-4000:])
     assert res.returncode == 0
     assert "POOL-GPU-OK" in res.stdout
+
+
+@pytest.mark.skipif(not torch.cuda.is_available(), reason="no GPU")
+def test_streamed_merge_two_ranks_one_gpu(tmp_path):
+    """Streamed merge plans with CUDA shards on an oversubscribed box:
+    the plan executor host-stages chunks (the same schedule the 8-GPU
+    RCCL run walks on-device); numerics vs the op-by-op reference."""
+    env = dict(os.environ)
+    env["LZY_AMD_STORAGE"] = str(tmp_path / "storage")
+    env["PYTHONPATH"] = str(ROOT) + os.pathsep + env.get("PYTHONPATH", "")
+    env.pop("RANK", None)
+    env.pop("WORLD_SIZE", None)
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    res = subprocess.run(
+        [
+            sys.executable, "-m", "torch.distributed.run",
+            "--nnodes=1", "--nproc-per-node=2",
+            "--master-addr", "127.0.0.1", "--master-port", str(port),
+            "tests/pool_script_streammerge.py",
+        ],
+        cwd=ROOT, env=env, capture_output=True, text=True, timeout=420,
+    )
+    if res.returncode != 0:
+        print("STDOUT:", res.stdout[-4000:])
+        print("STDERR:", res.stderr[-4000:])
+    assert res.returncode == 0
+    assert "STREAMMERGE-OK" in res.stdout
