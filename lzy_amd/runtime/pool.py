@@ -1750,6 +1750,11 @@ class _DriverScheduler:
         ):
             self.active_plan = None
             self.inflight -= 1
+            # survivors' plan_failed events arrive later as stale and are
+            # dropped — release their outstanding slots here
+            for r in ap["pending"]:
+                if r != rank and r in self.outstanding:
+                    self.outstanding[r] -= 1
             comp = self.components[ap["cid"]]
             for m in comp["order"]:
                 self.errors.append(LzyExecutionError(
