@@ -487,37 +487,37 @@ class WorkerAgent:
         return register
 
     def _flush_livelog(self, tid: str, final: bool = False) -> None:
+        # the whole read-compute-update-send runs under ONE lock: the
+        # periodic flusher and the op thread's final flush would
+        # otherwise race to ship the same delta twice (and could
+        # reorder chunks on the wire)
         with self._livelog_lock:
             rec = self._livelogs.get(tid)
             if rec is None:
                 return
             name, out_buf, err_buf, so, se = rec
 
-        def delta(buf, shipped):
-            s = buf.getvalue()[shipped:]
-            if not final:
-                # hold partial lines: the driver prints whole lines
-                cut = s.rfind("\n") + 1
-                s = s[:cut]
-            return s
+            def delta(buf, shipped):
+                s = buf.getvalue()[shipped:]
+                if not final:
+                    # hold partial lines: the driver prints whole lines
+                    cut = s.rfind("\n") + 1
+                    s = s[:cut]
+                return s
 
-        d_out = delta(out_buf, so)
-        d_err = delta(err_buf, se)
-        if not d_out and not d_err:
-            return
-        with self._livelog_lock:
-            rec = self._livelogs.get(tid)
-            if rec is None:
+            d_out = delta(out_buf, so)
+            d_err = delta(err_buf, se)
+            if not d_out and not d_err:
                 return
             rec[3] += len(d_out)
             rec[4] += len(d_err)
-        try:
-            self.ctrl.send_event({
-                "ev": "log_chunk", "task_id": tid, "name": name,
-                "rank": self.rank, "out": d_out, "err": d_err,
-            })
-        except (OSError, BrokenPipeError):
-            pass
+            try:
+                self.ctrl.send_event({
+                    "ev": "log_chunk", "task_id": tid, "name": name,
+                    "rank": self.rank, "out": d_out, "err": d_err,
+                })
+            except (OSError, BrokenPipeError):
+                pass
 
     def _livelog_flusher(self) -> None:
         while not self._shutdown:
@@ -773,6 +773,7 @@ class GpuPoolRuntime(Runtime):
         # different thread's active flight.
         self._flight = threading.Lock()
         self._flight_owner: Optional[str] = None
+        self._flight_guard = threading.Lock()  # atomic owner check+clear
         self._active_sched: Optional["_DriverScheduler"] = None
 
     @property
@@ -847,8 +848,15 @@ class GpuPoolRuntime(Runtime):
         try:
             self._finish_inner(workflow)
         finally:
-            if self._flight_owner == workflow.execution_id:
-                self._flight_owner = None
+            # atomic check-and-clear: a client-thread finish racing a
+            # side-thread abort must release exactly once — a double
+            # release could free the NEXT workflow's freshly acquired
+            # flight
+            with self._flight_guard:
+                owned = self._flight_owner == workflow.execution_id
+                if owned:
+                    self._flight_owner = None
+            if owned:
                 try:
                     self._flight.release()
                 except RuntimeError:
@@ -868,9 +876,12 @@ class GpuPoolRuntime(Runtime):
             produced = workflow.__dict__.get("_produced_entries", set())
             consumed = workflow.__dict__.get("_consumed_entries", set())
             self._drop_workflow_entries(workflow, keep=produced - consumed)
-            if self._journal is not None:
-                self._journal.close()
-                self._journal = None
+            # swap under the guard: finish and a racing abort must close
+            # the journal exactly once
+            with self._flight_guard:
+                j, self._journal = self._journal, None
+            if j is not None:
+                j.close()
 
     def abort(self, workflow: "LzyWorkflow") -> None:
         """Stop the graph, then finish.  When a scheduler batch is live
