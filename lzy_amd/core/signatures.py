@@ -135,7 +135,11 @@ def infer_call_signature(func: Callable, *args: Any, **kwargs: Any) -> CallSigna
 def check_type_compatible(value: Any, declared: Type) -> bool:
     """Lenient declared-vs-actual check (reference call.py:306-312): used
     to warn, not to block — lazy proxies and duck typing stay usable."""
-    if declared is object or declared is type(None) and value is None:
+    if declared is object:
+        return True
+    if value is None:
+        # Optional[T] unwraps to T for serializer lookup, so a legal
+        # None arrives here with declared=T — never warn on None
         return True
     try:
         return isinstance(value, declared)
