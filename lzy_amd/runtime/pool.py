@@ -151,6 +151,17 @@ class WorkerAgent:
                 target=self._livelog_flusher, daemon=True,
                 name=f"lzy-logflush-r{rank}",
             ).start()
+            # periodic liveness beacon (reference: AllocatorAgent
+            # heartbeat TimerTask, allocator-api AllocatorAgent.java:26).
+            # Socket EOF already detects process DEATH; heartbeats let
+            # the driver flag a HUNG-but-alive worker.
+            self._hb_period = float(
+                getattr(get_config(), "heartbeat_period_s", 2.0)
+            )
+            threading.Thread(
+                target=self._heartbeat, daemon=True,
+                name=f"lzy-hb-r{rank}",
+            ).start()
         OpLogCapture.instance().install()
 
     # -- serve loop ---------------------------------------------------------
@@ -527,6 +538,14 @@ class WorkerAgent:
             for tid in tids:
                 self._flush_livelog(tid)
 
+    def _heartbeat(self) -> None:
+        while not self._shutdown:
+            time.sleep(self._hb_period)
+            try:
+                self.ctrl.send_event({"ev": "heartbeat", "rank": self.rank})
+            except (OSError, BrokenPipeError):
+                return
+
     def _run_task(self, msg: dict) -> None:
         spec: TaskSpec = msg["spec"]
         self._settle(spec.wait_entries)
@@ -572,6 +591,9 @@ class GpuPool:
         self._ack_cv = threading.Condition()
         self._group_tags: Set[str] = set()
         self._seq = 0
+        # liveness: last event (heartbeat or otherwise) per rank
+        self.hb_last: Dict[int, float] = {}
+        self._stall_flagged: Set[int] = set()
 
     @classmethod
     def get(cls) -> "GpuPool":
@@ -637,6 +659,11 @@ class GpuPool:
 
         if self.is_driver:
             accept_thread.join()
+            if self.world > 1:
+                threading.Thread(
+                    target=self._stall_monitor, daemon=True,
+                    name="lzy-stallmon",
+                ).start()
             # short-circuit the loopback: rank 0 lives in this process,
             # so its commands and events skip the AF_UNIX socket (two
             # pickle hops saved per task/transfer/event).  The serve
@@ -663,6 +690,9 @@ class GpuPool:
 
     def _on_event(self, rank: int, msg: dict) -> None:
         ev = msg.get("ev")
+        self.hb_last[rank] = time.monotonic()  # any event proves liveness
+        if ev == "heartbeat":
+            return
         if ev in ("ack", "settled"):
             with self._ack_cv:
                 key = f"{ev}:{msg['tag']}"
@@ -706,6 +736,33 @@ class GpuPool:
             self.wait_acks("ack", tag, range(self.world))
             self._group_tags.add(tag)
         return tag
+
+    def _stall_monitor(self) -> None:
+        """Flag workers that stopped heartbeating but whose socket is
+        still open (hung in a collective, deadlocked op, stuck driver):
+        a warning + metric, never a kill — the reference's heartbeat-miss
+        policy (VM declared dead) is socket-EOF's job here; a hung-but-
+        alive rank still owns comms and killing it would poison them."""
+        from lzy_amd.config import get_config
+
+        period = float(getattr(get_config(), "heartbeat_period_s", 2.0))
+        while self.driver_ctrl is not None:
+            time.sleep(period)
+            now = time.monotonic()
+            for r in range(1, self.world):
+                last = self.hb_last.get(r)
+                if last is None:
+                    continue
+                silent = now - last
+                if silent > 4 * period and r not in self._stall_flagged:
+                    self._stall_flagged.add(r)
+                    METRICS.inc("lzy_worker_stalls")
+                    _LOG.warning(
+                        "worker rank %d silent for %.1fs (socket alive) — "
+                        "possibly hung", r, silent,
+                    )
+                elif silent <= 4 * period:
+                    self._stall_flagged.discard(r)
 
     def preflight(self) -> None:
         """Driver-side: run the communicator preflight on every rank
