@@ -47,6 +47,13 @@ _LOG = logging.getLogger("lzy_amd.treeplan")
 # driver side: component discovery + plan construction
 # ---------------------------------------------------------------------------
 
+def _gpu_count(call: Any) -> int:
+    try:
+        return int(call.env.provisioning.effective_gpu_count)
+    except AttributeError:
+        return 0
+
+
 def find_components(calls: Dict[str, Any]) -> List[List[str]]:
     """Connected components of pair_reduce tasks (edges: one task's output
     consumed by another), each topo-ordered.  Only components with >= 2
@@ -65,6 +72,8 @@ def find_components(calls: Dict[str, Any]) -> List[List[str]]:
         # the result cache (both lookup and write), and silently ignoring
         # @op(cache=True) would break the caching contract
         and not getattr(c, "cache", False)
+        # gang-provisioned ops keep gang semantics (op body on k ranks)
+        and _gpu_count(c) <= 1
     }
     if not members:
         return []

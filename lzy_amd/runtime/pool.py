@@ -32,7 +32,6 @@ import sys
 import tempfile
 import threading
 import time
-import uuid
 from typing import TYPE_CHECKING, Any, Dict, List, Optional, Sequence, Set, Tuple
 
 import torch
