@@ -386,6 +386,14 @@ def main() -> None:
             "stream_plans": int(METRICS.counter_value("lzy_stream_plans")),
             "stream_plan_ms": round(plan_stats.get("mean", 0.0) * 1e3, 3)
             if plan_stats.get("count") else None,
+            # mean per-rank recv wait inside plans: ~plan_ms on the
+            # critical rank means transfer-bound (expected); ~0 would
+            # mean compute-bound combines
+            "plan_recv_wait_ms": round(
+                METRICS.timing_stats("lzy_plan_recv_wait_s").get("mean", 0.0)
+                * 1e3, 3,
+            ) if METRICS.timing_stats("lzy_plan_recv_wait_s").get("count")
+            else None,
             "xgmi_gbps_est": xgmi_gbps_est,
             "stage_ms": stage_ms,
         },

@@ -33,6 +33,7 @@ from __future__ import annotations
 
 import logging
 import math
+import time
 from typing import Any, Dict, List, Optional, Sequence, Set, Tuple
 
 import torch
@@ -323,13 +324,18 @@ def run_stream_plan(
 
     recv_timeout = _dt.timedelta(seconds=wait_timeout)
 
+    stats = {"recv_wait_s": 0.0, "sends": 0, "recvs": 0}
+
     def wait_recv(w) -> None:
         # gloo honors the timeout (raises instead of hanging on a dead
         # peer); the RCCL path is stream-ordered — plain wait
+        t0 = time.perf_counter()
         if on_device:
             w.wait()
         else:
             w.wait(recv_timeout)
+        stats["recv_wait_s"] += time.perf_counter() - t0
+        stats["recvs"] += 1
 
     stream = STREAMS.next_stream()
     from contextlib import nullcontext
@@ -394,6 +400,7 @@ def run_stream_plan(
         staged: List[torch.Tensor] = []  # host staging keepalive
 
         def issue_send(t: torch.Tensor, dst: int) -> None:
+            stats["sends"] += 1
             if host_stage and t.is_cuda:
                 c = t.to("cpu")
                 staged.append(c)
@@ -466,4 +473,9 @@ def run_stream_plan(
             meta = describe_value(st["out"], value)
             results.append({**meta.to_wire(), "task": st["task"]})
         METRICS.inc("lzy_stream_plan_nodes", len(results))
+        # per-rank telemetry rides the first result (or a bare record):
+        # the driver aggregates recv-wait across ranks for the SCALE
+        # config block (overlap evidence)
+        if results:
+            results[0]["plan_stats"] = dict(stats)
     return results

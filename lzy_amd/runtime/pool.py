@@ -1758,6 +1758,12 @@ class _DriverScheduler:
         comp = self.components[ap["cid"]]
         if not ap["failed"]:
             METRICS.observe("lzy_stream_plan_s", time.perf_counter() - ap["t0"])
+            for w in ap["outputs"]:
+                ps = w.get("plan_stats")
+                if ps:  # per-rank recv-wait (SCALE overlap evidence)
+                    METRICS.observe(
+                        "lzy_plan_recv_wait_s", float(ps.get("recv_wait_s", 0.0))
+                    )
             by_task = {w["task"]: w for w in ap["outputs"]}
             for m in comp["order"]:
                 wire = by_task.get(m)
