@@ -7,9 +7,11 @@ combine rides inside the transfer's shadow and the pair-reduce finishes
 in ~transfer time.  On device the combine is the fused axpby kernel; on
 host it's a torch op.
 
-This is the building block for collapsing the DAG merge tree's latency
-(SURVEY §8 round-2 lever); exposed standalone so it is testable at any
-world size over gloo and 2-ranks-on-1-GPU.
+Round-2 note: the scheduler-integrated version of this idea is
+``channels/treeplan.py`` (whole merge TREES fold into one chunk-major
+plan with cross-level forwarding).  This standalone pair primitive
+remains the minimal testable form (chunked_transport_script) and the
+reference point the plan generalizes.
 """
 from __future__ import annotations
 

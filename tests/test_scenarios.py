@@ -89,7 +89,7 @@ _POOL_SCENARIOS = [
     "complex_graph", "repeated_ops_use_cache", "fully_cached_graph",
     "exec_fail", "cached_exception", "custom_serializer", "file_test",
     "nested_workflows", "whiteboards", "two_execution_one_wf",
-    "exception_serialize", "subprocess_with_startup",
+    "exception_serialize", "subprocess_with_startup", "stream_merge",
 ]
 
 
