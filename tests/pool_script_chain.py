@@ -52,7 +52,8 @@ def main() -> None:
         s = total(double(double(t)))
         assert float(s) == float(torch.arange(512).sum()) * 4, float(s)
     chains = METRICS.counter_value("lzy_chain_dispatches")
-    assert chains >= 2, f"expected chained dispatches, got {chains}"
+    if os.environ.get("LZY_CHAIN_DISPATCH", "1") not in ("0", "false"):
+        assert chains >= 2, f"expected chained dispatches, got {chains}"
 
     t0 = time.perf_counter()
     try:

@@ -74,7 +74,8 @@ def main() -> None:
         r_root, float(ref_root.sum()))
     assert abs(r_interior - float(ref01.sum())) < 1e-2
     plans = METRICS.counter_value("lzy_stream_plans")
-    assert plans >= 1, f"tree was not folded into a stream plan: {plans}"
+    if os.environ.get("LZY_STREAM_MERGE", "1") not in ("0", "false"):
+        assert plans >= 1, f"tree was not folded into a stream plan: {plans}"
     print(f"STREAMMERGE-PLANS {plans}", flush=True)
 
     # ---- alpha/beta variety: (1, -1) chain ------------------------------
