@@ -36,6 +36,10 @@ def total(t: torch.Tensor) -> float:
 
 
 def main() -> None:
+    if os.environ.get("LZY_STREAM_MERGE", "1") in ("0", "false"):
+        # no plans -> nothing to kill mid-plan; scenario vacuous
+        print("PLANDEATH-DETECTED (skipped: stream_merge off)", flush=True)
+        os._exit(0)
     os.environ.setdefault("LZY_SETTLE_WAIT_S", "15")
     if os.environ.get("RANK") == "2":
         # die exactly when this rank's plan schedule arrives
