@@ -104,11 +104,16 @@ class LzyWorkflow:
             self._finished = True
             LzyWorkflow._active.wf = None
             # values may outlive the workflow as plain tensors: guarantee
-            # their producing streams completed, then free the events
+            # their producing streams completed, then free the events.
+            # Iterate THIS workflow's entries — snapshot._values is the
+            # pool's shared store, which accumulates kept leaves across
+            # workflows (iterating it made exit O(total-ever-kept): a
+            # measurable per-DAG drift over thousands of workflows)
             from lzy_amd.runtime.streams import STREAMS
 
-            for eid in list(self.snapshot._values):
-                STREAMS.sync_and_drop(eid)
+            if self._snapshot is not None:
+                for eid in list(self._snapshot._entries):
+                    STREAMS.sync_and_drop(eid)
         return False
 
     # -- calls & barrier ------------------------------------------------------
