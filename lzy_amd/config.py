@@ -62,6 +62,7 @@ class Config:
     ack_wait_s: float = 300.0          # driver wait for worker acks
     # result cache / snapshot
     cache_enabled: bool = True
+    journal_ttl_hours: float = 168.0   # runtime prunes older journals at start (0 = off)
     # HBM store spill tier (pinned-host async)
     spill_enabled: bool = True
     spill_threshold_frac: float = 0.85

@@ -820,6 +820,19 @@ class GpuPoolRuntime(Runtime):
         self._journal_dir = journal_dir or os.path.join(
             tempfile.gettempdir(), "lzy_amd_journal"
         )
+        # opportunistic GC at runtime construction (reference: the
+        # lzy-service GarbageCollector sweeps stale executions
+        # periodically; one sweep per runtime keeps the journal dir
+        # bounded without a daemon)
+        try:
+            from lzy_amd.config import get_config
+            from lzy_amd.storage.gc import gc_journals
+
+            ttl_h = float(getattr(get_config(), "journal_ttl_hours", 168.0))
+            if ttl_h > 0:
+                gc_journals(self._journal_dir, ttl_seconds=ttl_h * 3600.0)
+        except Exception:  # noqa: BLE001 - GC must never block startup
+            pass
         self._journal: Optional[Journal] = None
         self._pool: Optional[GpuPool] = None
         # single-flight: the driver scheduler state is per-workflow.
