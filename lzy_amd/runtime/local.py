@@ -66,6 +66,21 @@ class LocalRuntime(Runtime):
         # its DB; nested workflows use their own Lzy()/runtime)
         import threading as _threading
 
+        # opportunistic journal GC (same sweep as GpuPoolRuntime)
+        try:
+            from lzy_amd.config import get_config
+            from lzy_amd.storage.gc import gc_journals
+
+            ttl_h = float(getattr(get_config(), "journal_ttl_hours", 168.0))
+            if ttl_h > 0:
+                gc_journals(
+                    self._journal_dir or os.path.join(
+                        tempfile.gettempdir(), "lzy_amd_journal"
+                    ),
+                    ttl_seconds=ttl_h * 3600.0,
+                )
+        except Exception:  # noqa: BLE001 - GC must never block startup
+            pass
         self._flight = _threading.Lock()
         # Lock has no ownership: remember WHICH workflow holds the flight
         # so finish/abort of a never-started (or already-finished) one
