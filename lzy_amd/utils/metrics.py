@@ -35,9 +35,18 @@ class Metrics:
         with self._lock:
             self._gauges[(name, tuple(sorted(labels.items())))] = value
 
+    # per-series cap: long-lived services observe forever; an unbounded
+    # list is a slow memory leak (~32 floats per workflow in the pool).
+    # When full, the series halves to its most recent half — stats
+    # become a recent-window view, which is what dashboards want anyway.
+    TIMING_CAP = 200_000
+
     def observe(self, name: str, seconds: float) -> None:
         with self._lock:
-            self._timings[name].append(seconds)
+            xs = self._timings[name]
+            xs.append(seconds)
+            if len(xs) > self.TIMING_CAP:
+                del xs[: len(xs) // 2]
 
     def timing_stats(self, name: str) -> Dict[str, float]:
         with self._lock:
