@@ -63,6 +63,7 @@ class Config:
     # result cache / snapshot
     cache_enabled: bool = True
     journal_ttl_hours: float = 168.0   # runtime prunes older journals at start (0 = off)
+    keep_hot_max: int = 10000          # post-exit leaf results kept hot (oldest dropped)
     # HBM store spill tier (pinned-host async)
     spill_enabled: bool = True
     spill_threshold_frac: float = 0.85

@@ -944,13 +944,43 @@ class GpuPoolRuntime(Runtime):
             # everything consumed is dropped for HBM pressure control
             produced = workflow.__dict__.get("_produced_entries", set())
             consumed = workflow.__dict__.get("_consumed_entries", set())
-            self._drop_workflow_entries(workflow, keep=produced - consumed)
+            keep = produced - consumed
+            self._drop_workflow_entries(workflow, keep=keep)
+            self._track_kept(keep)
             # swap under the guard: finish and a racing abort must close
             # the journal exactly once
             with self._flight_guard:
                 j, self._journal = self._journal, None
             if j is not None:
                 j.close()
+
+    def _track_kept(self, keep: Set[str]) -> None:
+        """Bound the post-exit leaf window.  Kept leaves stay hot for
+        post-exit reads; across MANY workflows they would accumulate
+        forever (one+ per workflow), so the oldest beyond keep_hot_max
+        are dropped pool-wide.  Reads of leaves older than the window
+        fall back to the durable tier (cache/whiteboard blobs) or raise —
+        the deliberate trade vs the reference's serialize-everything-
+        to-S3 hot path (see docs/architecture.md)."""
+        from collections import OrderedDict
+
+        from lzy_amd.config import get_config
+
+        pool = self.pool
+        kept: "OrderedDict[str, None]" = getattr(pool, "kept_hot", None)
+        if kept is None:
+            kept = pool.kept_hot = OrderedDict()
+        for eid in keep:
+            kept.pop(eid, None)
+            kept[eid] = None
+        cap = int(getattr(get_config(), "keep_hot_max", 10000))
+        if cap > 0 and len(kept) > cap:
+            evict = [kept.popitem(last=False)[0]
+                     for _ in range(len(kept) - cap // 2)]
+            pool.driver_ctrl.broadcast(
+                {"cmd": "drop_entries", "entries": evict}
+            )
+            METRICS.inc("lzy_kept_evicted", len(evict))
 
     def abort(self, workflow: "LzyWorkflow") -> None:
         """Stop the graph, then finish.  When a scheduler batch is live
