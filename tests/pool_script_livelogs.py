@@ -14,7 +14,7 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from lzy_amd import Lzy, op
 from lzy_amd.runtime.pool import GpuPool, GpuPoolRuntime
 
-OP_RUN_S = 2.0
+OP_RUN_S = 3.0
 
 
 class _Recorder(io.TextIOBase):

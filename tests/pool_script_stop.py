@@ -53,7 +53,7 @@ def main() -> None:
         with lzy.workflow("stop-fanout") as wf:
             th = threading.Thread(target=aborter, args=(wf,), daemon=True)
             th.start()
-            rs = [slow(i) for i in range(12)]
+            rs = [slow(i) for i in range(16)]
             vals = [int(r) for r in rs]  # blocks in the barrier
     except WorkflowAbortedError:
         got_abort = True
@@ -63,9 +63,10 @@ def main() -> None:
     assert got_abort, "barrier did not raise WorkflowAbortedError"
     assert aborted_at.get("raised"), "side-thread abort did not raise"
     latency = t_raised - aborted_at["t"]
-    # 12 ops x 1.2 s on 4 executor slots = 3 waves (~3.6 s) if the queue
-    # drains; prompt cancel bounds it to ~one wave + overhead
-    assert latency < 2.5 * OP_SLEEP, f"teardown too slow: {latency:.2f}s"
+    # 16 ops x 1.2 s on 4 executor slots = 4 waves (~4.8 s) if the queue
+    # drains; prompt cancel bounds it to ~one wave + overhead (margin
+    # sized for loaded CI boxes)
+    assert latency < 2.7 * OP_SLEEP, f"teardown too slow: {latency:.2f}s"
     print(f"STOP-LATENCY {latency:.2f}", flush=True)
 
     # pool must remain usable after the stop
