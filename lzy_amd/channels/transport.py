@@ -282,11 +282,16 @@ class Transport:
     # their xGMI links instead of serializing on the comm stream.
 
     def send_ops(self, value: Any, prepickled: Optional[bytes], dst: int,
-                 offset_chunks: int = 0):
+                 offset_chunks: int = 0, tag: int = 0):
         """Build send P2POps; returns (ops, keepalive).
 
         ``offset_chunks`` resumes a partially completed transfer: the
         first ``offset_chunks`` chunks are assumed delivered and skipped.
+        ``tag``: driver-assigned per-transfer tag.  gloo matches p2p by
+        (pair, tag, order); unique tags mean the stale posted recvs of a
+        FAILED transfer can never swallow a later transfer's chunks.
+        NCCL/RCCL ignores tags — that path keeps the strict
+        driver-sequenced issue-order discipline instead.
         """
         if isinstance(value, torch.Tensor):
             t = value.detach()
@@ -307,7 +312,8 @@ class Transport:
                 chunks = [t]
             else:
                 chunks = self._chunks(t.view(-1), per, offset_chunks)
-            ops = [dist.P2POp(dist.isend, c, dst, group=self._pg) for c in chunks]
+            ops = [dist.P2POp(dist.isend, c, dst, group=self._pg, tag=tag)
+                   for c in chunks]
             return ops, t
         data = prepickled if prepickled is not None else pickle_value(value)
         buf = torch.frombuffer(bytearray(data), dtype=torch.uint8)
@@ -316,11 +322,12 @@ class Transport:
             chunks = [buf]
         else:
             chunks = self._chunks(buf, per, offset_chunks)
-        ops = [dist.P2POp(dist.isend, c, dst, group=self._pg) for c in chunks]
+        ops = [dist.P2POp(dist.isend, c, dst, group=self._pg, tag=tag)
+               for c in chunks]
         return ops, buf
 
     def recv_ops(self, meta: EntryMeta, src: int, offset_chunks: int = 0,
-                 into: Optional[torch.Tensor] = None):
+                 into: Optional[torch.Tensor] = None, tag: int = 0):
         """Build recv P2POps; returns (ops, finalize) where finalize() ->
         the received value (call after the issued works complete).
 
@@ -351,7 +358,8 @@ class Transport:
                 chunks = [buf]
             else:
                 chunks = self._chunks(buf.view(-1), per, offset_chunks)
-            ops = [dist.P2POp(dist.irecv, c, src, group=self._pg) for c in chunks]
+            ops = [dist.P2POp(dist.irecv, c, src, group=self._pg, tag=tag)
+                   for c in chunks]
 
             def finalize():
                 out = buf
@@ -368,7 +376,8 @@ class Transport:
             chunks = [buf]
         else:
             chunks = self._chunks(buf, per, offset_chunks)
-        ops = [dist.P2POp(dist.irecv, c, src, group=self._pg) for c in chunks]
+        ops = [dist.P2POp(dist.irecv, c, src, group=self._pg, tag=tag)
+               for c in chunks]
         return ops, (lambda: unpickle_value(buf.numpy().tobytes()))
 
     @staticmethod
@@ -381,15 +390,15 @@ class Transport:
     # -- single-transfer conveniences ----------------------------------------
 
     def isend_value(self, value: Any, prepickled: Optional[bytes], dst: int,
-                    offset_chunks: int = 0):
+                    offset_chunks: int = 0, tag: int = 0):
         """Issue non-blocking send(s); returns (works, keepalive)."""
-        ops, keep = self.send_ops(value, prepickled, dst, offset_chunks)
+        ops, keep = self.send_ops(value, prepickled, dst, offset_chunks, tag)
         return self.issue(ops), keep
 
     def irecv_value(self, meta: EntryMeta, src: int, offset_chunks: int = 0,
-                    into: Optional[torch.Tensor] = None):
+                    into: Optional[torch.Tensor] = None, tag: int = 0):
         """Issue non-blocking recv; returns (works, finalize)."""
-        ops, fin = self.recv_ops(meta, src, offset_chunks, into)
+        ops, fin = self.recv_ops(meta, src, offset_chunks, into, tag)
         return self.issue(ops), fin
 
     @staticmethod

@@ -224,7 +224,9 @@ def build_plan(
                 issue_topo = member_order.index(child)
                 for st in steps_by_rank[src_rank]:
                     if st.get("task") == child:
-                        st["send_to"].append(my_rank)
+                        # (destination rank, consumer topo): the tag the
+                        # receiver posts is derived from ITS node's topo
+                        st["send_to"].append([my_rank, topo_idx])
                         break
             pair_edges.setdefault((src_rank, my_rank), []).append(
                 (issue_topo, topo_idx)
@@ -262,6 +264,10 @@ def build_plan(
         per = max((1 << 20) // elem, math.ceil(numel / 4))
     return {
         "plan_id": plan_id,
+        # per-plan tag base: edge tag = (nonce + topo) so a FAILED
+        # plan's stale posted recvs on pg_stream can never match a later
+        # plan's chunks (gloo matches tags; RCCL ignores them)
+        "nonce": abs(hash(plan_id)) % (1 << 20) << 9,
         "shape": list(shape),
         "dtype": dtype,
         "numel": numel,
@@ -382,6 +388,11 @@ def run_stream_plan(
             s = j * per
             return s, min(s + per, numel)
 
+        nonce = plan.get("nonce", 0)
+
+        def edge_tag(topo: int) -> int:
+            return (nonce + topo) % (1 << 30)
+
         def post_up_to(src: int, upto_flat_idx: int) -> None:
             sts = seq_by_src[src]
             width = len(sts)
@@ -392,21 +403,25 @@ def run_stream_plan(
                 st = sts[st]
                 s, e = chunk_bounds(j)
                 recv_works[(st["topo"], j)] = dist.irecv(
-                    recv_buf[st["topo"]][s:e], src=src, group=pg
+                    recv_buf[st["topo"]][s:e], src=src, group=pg,
+                    tag=edge_tag(st["topo"]),
                 )
                 posted[src] += 1
 
         send_works: List[Any] = []
         staged: List[torch.Tensor] = []  # host staging keepalive
 
-        def issue_send(t: torch.Tensor, dst: int) -> None:
+        def issue_send(t: torch.Tensor, dst: int, topo: int) -> None:
             stats["sends"] += 1
+            tg = edge_tag(topo)
             if host_stage and t.is_cuda:
                 c = t.to("cpu")
                 staged.append(c)
-                send_works.append(dist.isend(c, dst=dst, group=pg))
+                send_works.append(dist.isend(c, dst=dst, group=pg, tag=tg))
             else:
-                send_works.append(dist.isend(t.contiguous(), dst=dst, group=pg))
+                send_works.append(
+                    dist.isend(t.contiguous(), dst=dst, group=pg, tag=tg)
+                )
 
         try:
             from lzy_amd import ops as _ops
@@ -431,7 +446,8 @@ def run_stream_plan(
             s, e = chunk_bounds(j)
             for st in my_steps:
                 if st["op"] == "leaf_send":
-                    issue_send(leaf_flat[st["topo"]][s:e], st["dst"])
+                    issue_send(leaf_flat[st["topo"]][s:e], st["dst"],
+                               st["topo"])
                     continue
                 a = (
                     local_flat[st["topo"]]
@@ -455,8 +471,8 @@ def run_stream_plan(
                         if st["local2_is_entry"] else outs[st["local2"]]
                     )
                     combine(out[s:e], a[s:e], b[s:e], st["alpha"], st["beta"])
-                for dst in st["send_to"]:
-                    issue_send(out[s:e], dst)
+                for dst, fwd_topo in st["send_to"]:
+                    issue_send(out[s:e], dst, fwd_topo)
 
         for w in send_works:
             w.wait()

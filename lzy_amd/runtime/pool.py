@@ -195,28 +195,47 @@ class WorkerAgent:
         elif cmd == "xfer_send_batch":
             ops_all: list = []
             keeps: list = []
+            failed_items: list = []
             for item in msg["items"]:
                 eid = item["entry"]
-                if not self.store.has(eid):  # defense: settle may be landing
-                    # config-derived (settle_wait_s, cached by the store):
-                    # a multi-GB fan-in over one xGMI link can
-                    # legitimately outlast a fixed small bound
-                    self.store.wait_present(eid)
-                value = self.store.get(eid)
+                try:
+                    if not self.store.has(eid):  # settle may be landing
+                        # config-derived (settle_wait_s, cached by the
+                        # store): a multi-GB fan-in over one xGMI link
+                        # can legitimately outlast a fixed small bound
+                        self.store.wait_present(eid)
+                    value = self.store.get(eid)
+                except (KeyError, RuntimeError) as e:
+                    # this rank never (or no longer) holds the entry —
+                    # e.g. its own inbound copy died with the source.
+                    # Tell the driver so it can poison the waiting
+                    # consumer instead of letting it block on a recv
+                    # that will never be fed.
+                    failed_items.append((item, f"{type(e).__name__}: {e}"))
+                    continue
                 ops, keep = self.transport.send_ops(
-                    value, self.store.pickled.get(eid), item["dst"]
+                    value, self.store.pickled.get(eid), item["dst"],
+                    tag=item.get("tag", 0),
                 )
                 ops_all += ops
                 keeps.append(keep)
             works = Transport.issue(ops_all)
             self._outbox.append((works, keeps))
             self._prune_outbox()
+            for item, err in failed_items:
+                self.ctrl.send_event({
+                    "ev": "xfer_failed", "rank": self.rank,
+                    "entry": item["entry"], "dst": item["dst"],
+                    "error": err,
+                })
         elif cmd == "xfer_recv_batch":
             ops_all = []
             fins: list = []
             for item in msg["items"]:
                 meta = EntryMeta.from_wire(item["meta"])
-                ops, fin = self.transport.recv_ops(meta, item["src"])
+                ops, fin = self.transport.recv_ops(
+                    meta, item["src"], tag=item.get("tag", 0)
+                )
                 fins.append((item["entry"], len(ops), fin))
                 ops_all += ops
             works = Transport.issue(ops_all)
@@ -461,8 +480,19 @@ class WorkerAgent:
                 continue
             works, fin = pending
             if works:
+                # gloo waits are bounded: a transfer whose sender died or
+                # never sent must fail this task promptly, not hang the
+                # executor forever.  RCCL waits are stream-enqueue
+                # (non-host-blocking) — no timeout semantics needed.
+                import datetime as _dt
+
+                bounded = not self.transport._cuda_p2p
+                tmo = _dt.timedelta(seconds=self.store._settle_wait)
                 for w in works:
-                    w.wait()
+                    if bounded:
+                        w.wait(tmo)
+                    else:
+                        w.wait()
             value = fin()
             pickled = None
             if not isinstance(value, torch.Tensor):
@@ -593,6 +623,7 @@ class GpuPool:
         # liveness: last event (heartbeat or otherwise) per rank
         self.hb_last: Dict[int, float] = {}
         self._stall_flagged: Set[int] = set()
+        self.dead_ranks: Set[int] = set()
 
     @classmethod
     def get(cls) -> "GpuPool":
@@ -692,6 +723,8 @@ class GpuPool:
         self.hb_last[rank] = time.monotonic()  # any event proves liveness
         if ev == "heartbeat":
             return
+        if ev == "worker_lost":
+            self.dead_ranks.add(rank)
         if ev in ("ack", "settled"):
             with self._ack_cv:
                 key = f"{ev}:{msg['tag']}"
@@ -749,6 +782,8 @@ class GpuPool:
             time.sleep(period)
             now = time.monotonic()
             for r in range(1, self.world):
+                if r in self.dead_ranks:
+                    continue  # dead, not hung — already handled
                 last = self.hb_last.get(r)
                 if last is None:
                     continue
@@ -1021,14 +1056,17 @@ class GpuPoolRuntime(Runtime):
         owner = next(iter(meta.owners - {0}), None)
         if owner is None:
             return
+        xtag = pool.next_seq() % (1 << 30)
         pool.driver_ctrl.send(
             owner,
-            {"cmd": "xfer_send_batch", "items": [{"entry": entry_id, "dst": 0}]},
+            {"cmd": "xfer_send_batch",
+             "items": [{"entry": entry_id, "dst": 0, "tag": xtag}]},
         )
         pool.driver_ctrl.send(
             0,
             {"cmd": "xfer_recv_batch",
-             "items": [{"entry": entry_id, "src": owner, "meta": meta.to_wire()}]},
+             "items": [{"entry": entry_id, "src": owner,
+                        "meta": meta.to_wire(), "tag": xtag}]},
         )
         tag = f"f{pool.next_seq()}"
         pool.driver_ctrl.send(0, {"cmd": "settle", "entries": [entry_id], "tag": tag})
@@ -1315,19 +1353,33 @@ class _DriverScheduler:
                         else:
                             self.journal.record(ct, "cancelled")
             elif ev == "agent_error":
-                self.errors.append(
-                    LzyExecutionError(f"agent rank {rank}: {msg['error']}")
-                )
                 tid = msg.get("task_id")
+                if (
+                    tid is not None
+                    and tid in self.calls
+                    and tid not in self.task_dispatch_ts
+                ):
+                    # stale: the task was already terminally accounted
+                    # (or RETRIED) — its late settle-timeout error must
+                    # not fail a workflow that recovered
+                    continue
                 if tid is None or tid not in self.task_dispatch_ts:
                     # serve-loop failure on a non-task command (transfer
                     # staging etc.): the waiting task surfaces its own
                     # failure later — decrementing inflight here would
                     # underflow and leak its completion event into the
                     # next batch's scheduler
+                    self.errors.append(
+                        LzyExecutionError(f"agent rank {rank}: {msg['error']}")
+                    )
                     continue
                 if rank in self.outstanding:
                     self.outstanding[rank] -= 1
+                if self._retry_after_agent_error(rank, tid, msg):
+                    continue
+                self.errors.append(
+                    LzyExecutionError(f"agent rank {rank}: {msg['error']}")
+                )
                 if tid not in failed_tasks:
                     failed_tasks.add(tid)
                     self.journal.record(tid, "failed", msg["error"])
@@ -1368,6 +1420,31 @@ class _DriverScheduler:
                 self.task_dispatch_ts.pop(tid, None)
                 self.chained_waits.pop(tid, None)
                 self.journal.record(tid, "cancelled", self.stopping or "")
+            elif ev == "xfer_failed":
+                # a source rank could not provide an entry it was asked
+                # to send: purge the target's claimed ownership and fail
+                # (or retry, via the agent_error machinery) every task on
+                # the destination waiting for it
+                eid, dst = msg["entry"], msg["dst"]
+                m = self.meta.get(eid)
+                if m is not None:
+                    m.owners.discard(dst)
+                self.transferred_now.discard((dst, eid))
+                for tid in list(self.task_dispatch_ts):
+                    call = self.calls.get(tid)
+                    if (
+                        call is None
+                        or dst not in self.task_ranks.get(tid, ())
+                        or eid not in call.input_entry_ids()
+                    ):
+                        continue
+                    pool.events.put((dst, {
+                        "ev": "agent_error", "task_id": tid,
+                        "error": (
+                            f"transfer of {eid} from rank {rank} failed: "
+                            f"{msg.get('error', '')}"
+                        ),
+                    }))
             elif ev == "stop":
                 if msg.get("sched") == id(self):
                     self._initiate_stop(dag, msg.get("reason", "stopped"))
@@ -1482,11 +1559,17 @@ class _DriverScheduler:
                         ipc_imports.append((r, eid))
                         METRICS.inc("lzy_transfers_ipc")
                     else:
+                        # unique per-transfer tag: gloo matches (pair,
+                        # tag, order), so a FAILED transfer's stale
+                        # posted recvs can never swallow a later
+                        # transfer's chunks on the same pair
+                        xtag = pool.next_seq() % (1 << 30)
                         sends_by_owner.setdefault(owner, []).append(
-                            {"entry": eid, "dst": r}
+                            {"entry": eid, "dst": r, "tag": xtag}
                         )
                         recvs_by_rank.setdefault(r, []).append(
-                            {"entry": eid, "src": owner, "meta": meta.to_wire()}
+                            {"entry": eid, "src": owner,
+                             "meta": meta.to_wire(), "tag": xtag}
                         )
                         METRICS.inc("lzy_transfers")
                     wait_entries_per_rank[r].append(eid)
@@ -1842,6 +1925,74 @@ class _DriverScheduler:
             if not nxt_comp["launched"] and not nxt_comp["broken"]:
                 self._launch_plan(nxt_cid, nxt_comp)
 
+    def _retry_after_agent_error(self, rank: int, tid: str, msg: dict) -> bool:
+        """A task's settle failed (typically: its input's source rank
+        died mid-transfer, poisoning or timing out the wait).  If the
+        retry budget allows and every input is still recoverable, account
+        the attempt as terminated and re-dispatch instead of failing the
+        workflow (the same failover worker death itself gets)."""
+        call = self.calls.get(tid)
+        if (
+            call is None
+            or self.retry_budget.get(tid, self._max_retries) <= 0
+            or call.env.provisioning.effective_gpu_count > 1
+            or tid in self.gang_pending
+            or self.stopping is not None
+            or not self.outstanding
+        ):
+            _LOG.warning("agent-error retry declined for %s (budget=%s)",
+                         getattr(call, "callable_name", tid),
+                         self.retry_budget.get(tid, self._max_retries))
+            return False
+        # recoverability uses the same machinery as worker death; no
+        # concurrent sweep is running, so retrying={tid} only
+        self.retrying = {tid}
+        ok = self._recover_inputs(call, prefer_reliable=True)
+        self.retrying = set()
+        if not ok:
+            _LOG.warning("agent-error retry: inputs unrecoverable for %s",
+                         call.callable_name)
+            return False
+        self.retry_budget[tid] = self.retry_budget.get(tid, self._max_retries) - 1
+        self.inflight -= 1
+        self.task_dispatch_ts.pop(tid, None)
+        self.chained_waits.pop(tid, None)
+        self.task_ranks.pop(tid, None)
+        self.dispatched.discard(tid)
+        self.journal.record(tid, "retry", f"agent error on rank {rank}")
+        METRICS.inc("lzy_task_retries")
+        _LOG.warning(
+            "task %s failed on rank %d (%s); re-dispatching",
+            call.callable_name, rank, msg.get("error", ""),
+        )
+        if all(d in self.dag_completed for d in self.task_deps.get(tid, ())):
+            self._dispatch(tid)
+        return True
+
+    def _reroot_on_driver(self, eid: str, meta, snap, store) -> bool:
+        """Make the driver a confirmed owner of ``eid`` from its own
+        store or the durable tier; True on success."""
+        if store.has(eid):
+            meta.owners.add(0)
+            return True
+        try:
+            entry = snap.get_entry(eid)
+        except KeyError:
+            _LOG.warning("reroot: no snapshot entry for %s", eid)
+            return False
+        if not snap.storage.blob_exists(entry.storage_uri):
+            _LOG.warning("reroot: no durable blob at %s", entry.storage_uri)
+            return False
+        value = snap.load(eid)  # lands in the rank-0 store (shared dict)
+        fresh = describe_value(eid, value)
+        fresh.owners = {0}
+        if fresh.kind == KIND_BYTES:
+            data = pickle_value(value)
+            store.pickled[eid] = data
+            fresh.nbytes = len(data)
+        self.meta[eid] = fresh
+        return True
+
     # -- worker death --------------------------------------------------------
 
     def _on_worker_lost(self, rank: int, dag, failed_tasks: Set[str]) -> None:
@@ -1956,10 +2107,16 @@ class _DriverScheduler:
                 self._dispatch(tid)
         self.retrying = set()
 
-    def _recover_inputs(self, call: "LzyCall") -> bool:
+    def _recover_inputs(self, call: "LzyCall",
+                        prefer_reliable: bool = False) -> bool:
         """Check every input of a to-be-retried task is reachable from a
         surviving rank — restoring driver ownership from the durable tier
-        where needed.  Returns False when any input is gone for good."""
+        where needed.  Returns False when any input is gone for good.
+
+        ``prefer_reliable`` (agent-error retries): re-root the input on
+        the driver/durable tier even when a remote owner LOOKS alive —
+        the failure may have raced ahead of that owner's death event, and
+        a retry that trusts it just burns the budget."""
         snap = self.workflow.snapshot
         store = self.pool.agent.store
         for eid in call.input_entry_ids():
@@ -1972,7 +2129,16 @@ class _DriverScheduler:
                 if store.has(eid) or snap.has_value(eid):
                     continue
                 return False
-            if any(o in self.outstanding for o in meta.owners):
+            if prefer_reliable and self._reroot_on_driver(eid, meta, snap, store):
+                continue
+            # only CONFIRMED live owners count: a rank in owners whose
+            # copy was merely INITIATED (transferred_now) may never have
+            # received it — e.g. the source died mid-send
+            if any(
+                o in self.outstanding
+                and (o, eid) not in self.transferred_now
+                for o in meta.owners
+            ):
                 continue
             if store.has(eid):
                 meta.owners.add(0)

@@ -324,6 +324,10 @@ def run_taskspec(
             # blob's existence, so the format must already be readable
             _write_fmt(storage, out_uris[eid], fmt)
             storage.write_bytes(out_uris[eid], data)
+            # the driver must learn the durable location on the WRITE
+            # path too (not only on later cache hits): failover re-roots
+            # lost inputs from this blob (the channel's storage peer)
+            wire["uri"] = out_uris[eid]
 
     METRICS.inc("lzy_op_runs", op=spec.name)
     elapsed = time.perf_counter() - t0

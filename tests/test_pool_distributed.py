@@ -119,6 +119,43 @@ def test_streamed_merge_tree_world4(tmp_path):
     assert "STREAMMERGE-OK" in r.stdout
 
 
+def test_transfer_source_death_durable_failover(tmp_path):
+    """A transfer's SOURCE rank dies mid-send; the consumer's settle
+    times out and the driver re-sources the cached input from the
+    durable tier, re-dispatching instead of failing (reference:
+    transferFailed -> storage peer, SlotsService.java:191-240)."""
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    base_env = dict(os.environ)
+    base_env["LZY_AMD_STORAGE"] = str(tmp_path / "storage")
+    base_env["PYTHONPATH"] = str(ROOT) + os.pathsep + base_env.get("PYTHONPATH", "")
+    base_env.update(
+        MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port), WORLD_SIZE="3"
+    )
+    procs = []
+    for rank in range(3):
+        env = dict(base_env)
+        env["RANK"] = str(rank)
+        env["LOCAL_RANK"] = str(rank)
+        procs.append(subprocess.Popen(
+            [sys.executable, "tests/pool_script_xferdeath.py"],
+            cwd=ROOT, env=env, stdout=subprocess.PIPE, stderr=subprocess.PIPE,
+            text=True,
+        ))
+    out0, err0 = procs[0].communicate(timeout=180)
+    for p in procs[1:]:
+        try:
+            p.wait(timeout=60)
+        except subprocess.TimeoutExpired:
+            p.kill()
+    assert "XFERDEATH-RECOVERED" in out0, out0[-3000:] + err0[-2000:]
+    # the injected death must actually have fired (rank 1 exits 9)
+    assert procs[1].returncode == 9, procs[1].returncode
+
+
 def test_worker_death_during_stream_plan(tmp_path):
     """A participant dying mid-plan fails the folded tasks promptly and
     raises on the client — no hang, survivors' stale events dropped."""

@@ -122,9 +122,10 @@ def test_build_plan_8_leaves_pipelines_across_levels():
     assert nr[root] == 0
     r0_nodes = [s for s in plan["steps_by_rank"][0] if s["op"] == "node"]
     assert len(r0_nodes) == 3  # level0, level1, root
-    # forwarding: rank 2's level-1 node sends to rank 0 (the root)
+    # forwarding: rank 2's level-1 node sends to rank 0 (the root);
+    # send_to carries (dst, consumer_topo) for tag derivation
     r2_nodes = [s for s in plan["steps_by_rank"][2] if s["op"] == "node"]
-    assert any(0 in s["send_to"] for s in r2_nodes)
+    assert any(any(d == 0 for d, _ in s["send_to"]) for s in r2_nodes)
     # pure senders only leaf_send
     r7 = plan["steps_by_rank"][7]
     assert all(s["op"] == "leaf_send" for s in r7)
