@@ -815,9 +815,10 @@ class GpuPool:
         tag = f"b{self.next_seq()}"
         self.driver_ctrl.broadcast({"cmd": "barrier", "tag": tag})
         deadline = time.monotonic() + 600
+        want = self.world - len(self.dead_ranks)
         ts: Dict[int, float] = {}
         stash = []
-        while len(ts) < self.world:
+        while len(ts) < want:
             if time.monotonic() > deadline:
                 raise TimeoutError("pool barrier timed out")
             try:
@@ -1085,7 +1086,12 @@ class _DriverScheduler:
         self.journal = journal
         self.meta: Dict[str, EntryMeta] = getattr(workflow, "_entry_meta", {})
         workflow._entry_meta = self.meta
-        self.outstanding: Dict[int, int] = {r: 0 for r in range(pool.world)}
+        # dead ranks stay excluded across batches (a later workflow must
+        # not dispatch to a corpse)
+        self.outstanding: Dict[int, int] = {
+            r: 0 for r in range(pool.world)
+            if r not in getattr(pool, "dead_ranks", set())
+        }
         self.task_ranks: Dict[str, List[int]] = {}
         self.task_dispatch_ts: Dict[str, float] = {}
         self.gang_pending: Dict[str, Set[int]] = {}

@@ -35,6 +35,11 @@ def main() -> None:
             rs = [maybe_die(i) for i in range(6)]
             vals = [int(r) for r in rs]
         assert vals == [i * 10 for i in range(6)], vals
+        # the pool must stay usable AFTER the death: a later workflow
+        # schedules only onto surviving ranks
+        with lzy.workflow("death-after"):
+            vals2 = [int(r) for r in (maybe_die(10), maybe_die(11))]
+        assert vals2 == [100, 110], vals2
         print("DEATH-RECOVERED", flush=True)
     except LzyExecutionError as e:
         print(f"DEATH-DETECTED: {e}", flush=True)
