@@ -752,6 +752,9 @@ class GpuPool:
             self._acks.pop(key, None)
             return self._ack_payloads.pop(key, {})
 
+    def live_ranks(self):
+        return [r for r in range(self.world) if r not in self.dead_ranks]
+
     def next_seq(self) -> int:
         self._seq += 1
         return self._seq
@@ -764,6 +767,15 @@ class GpuPool:
             return "__default__"
         tag = ",".join(map(str, ranks))
         if tag not in self._group_tags:
+            if self.dead_ranks:
+                # dist.new_group is collective over the DEFAULT group —
+                # a dead rank can never join, so creation would hang.
+                # Cached groups keep working; new ones cannot exist.
+                raise RuntimeError(
+                    f"cannot create process group {ranks}: rank(s) "
+                    f"{sorted(self.dead_ranks)} are dead (new_group is "
+                    "collective over all ranks)"
+                )
             self.driver_ctrl.broadcast({"cmd": "new_group", "ranks": ranks, "tag": tag})
             self.wait_acks("ack", tag, range(self.world))
             self._group_tags.add(tag)
@@ -807,7 +819,7 @@ class GpuPool:
             return
         tag = f"pf{self.next_seq()}"
         self.driver_ctrl.broadcast({"cmd": "preflight", "tag": tag})
-        self.wait_acks("ack", tag, range(self.world), timeout=240.0)
+        self.wait_acks("ack", tag, self.live_ranks(), timeout=240.0)
 
     def sync_all(self) -> Dict[int, float]:
         """Barrier across all ranks (through exec queues, so it orders after
@@ -923,7 +935,7 @@ class GpuPoolRuntime(Runtime):
             pool.driver_ctrl.broadcast(
                 {"cmd": "load_serializers", "payload": payload, "tag": tag}
             )
-            pool.wait_acks("ack", tag, range(pool.world))
+            pool.wait_acks("ack", tag, pool.live_ranks())
             pool._last_ser_payload = payload
 
     def exec(self, workflow: "LzyWorkflow", calls: Sequence["LzyCall"]) -> None:
@@ -1496,7 +1508,15 @@ class _DriverScheduler:
             if ranks is None:
                 self.deferred_gangs.append(task_id)
                 return
-            tag = pool.ensure_group(ranks)
+            try:
+                tag = pool.ensure_group(ranks)
+            except RuntimeError as e:
+                # post-death pools cannot create NEW groups (collective
+                # over all ranks): fail the gang op with a typed error
+                # instead of crashing the scheduler loop
+                self.errors.append(BadProvisioningError(str(e)))
+                self.journal.record(task_id, "failed", str(e))
+                return
             gang = {"ranks": ranks, "tag": tag}
         else:
             ranks = [self._pick_rank(call)]
