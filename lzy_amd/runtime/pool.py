@@ -1125,6 +1125,12 @@ class _DriverScheduler:
         # condition waits give correct ordering) — the driver round-trip
         # leaves the dependency-chain critical path
         self.task_deps: Dict[str, List[str]] = {}
+        self.task_children: Dict[str, List[str]] = {}
+        # chain-eligibility worklist: a task is (re)examined only when
+        # one of its producers gets dispatched — scanning ALL calls per
+        # completion was O(n^2) on wide graphs (measured 0.27 -> 0.54
+        # ms/op from width 128 to 1024)
+        self._chain_candidates: Set[str] = set()
         self.entry_producer: Dict[str, str] = {}
         self.dispatched: Set[str] = set()
         self.dag_completed: Set[str] = set()
@@ -1236,6 +1242,8 @@ class _DriverScheduler:
             })
             dag.add_task(c.id, deps)
             self.task_deps[c.id] = deps
+            for d in deps:
+                self.task_children.setdefault(d, []).append(c.id)
         self.entry_producer = producer
         dag.seal()
         self._discover_components()
@@ -1666,6 +1674,7 @@ class _DriverScheduler:
         if chain_rank is not None and pending_local:
             self.chained_waits[task_id] = (chain_rank, sorted(pending_local))
         self.task_dispatch_ts[task_id] = dispatch_t0
+        self._chain_candidates.update(self.task_children.get(task_id, ()))
         self.journal.record(task_id, "scheduled", call.callable_name)
         METRICS.observe("lzy_dispatch", time.perf_counter() - dispatch_t0)
 
@@ -2192,8 +2201,15 @@ class _DriverScheduler:
         rank, without waiting for their completion events."""
         if not self._chain_enabled or self.stopping is not None:
             return
-        for tid, call in self.calls.items():
-            if tid in self.dispatched:
+        while self._chain_candidates:
+            cands = self._chain_candidates
+            self._chain_candidates = set()
+            self._try_chain_over(cands)
+
+    def _try_chain_over(self, cands: Set[str]) -> None:
+        for tid in sorted(cands):
+            call = self.calls.get(tid)
+            if call is None or tid in self.dispatched:
                 continue
             if call.env.provisioning.effective_gpu_count > 1:
                 continue  # gangs stay on the completion-driven path
