@@ -206,6 +206,39 @@ def test_concurrent_gangs_world8(tmp_path):
     assert "GANGS-OK" in r.stdout
 
 
+def test_sigint_mid_barrier(tmp_path):
+    """Ctrl-C in the blocked barrier stops the graph promptly and the
+    interrupt propagates (queued ops cancelled, not drained)."""
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    base_env = dict(os.environ)
+    base_env["LZY_AMD_STORAGE"] = str(tmp_path / "storage")
+    base_env["PYTHONPATH"] = str(ROOT) + os.pathsep + base_env.get("PYTHONPATH", "")
+    base_env.update(
+        MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port), WORLD_SIZE="2"
+    )
+    procs = []
+    for rank in (0, 1):
+        env = dict(base_env)
+        env["RANK"] = str(rank)
+        env["LOCAL_RANK"] = str(rank)
+        procs.append(subprocess.Popen(
+            [sys.executable, "tests/pool_script_sigint.py"],
+            cwd=ROOT, env=env, stdout=subprocess.PIPE, stderr=subprocess.PIPE,
+            text=True,
+        ))
+    out0, err0 = procs[0].communicate(timeout=120)
+    for p in procs[1:]:
+        try:
+            p.wait(timeout=30)
+        except subprocess.TimeoutExpired:
+            p.kill()
+    assert "SIGINT-HANDLED" in out0, out0[-2000:] + err0[-2000:]
+
+
 def test_stop_graph_mid_flight(tmp_path):
     """Abort of a slow fan-out mid-barrier: queued tasks cancelled on all
     ranks, barrier raises promptly, pool stays usable (StopGraph)."""
