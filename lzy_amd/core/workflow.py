@@ -146,6 +146,13 @@ class LzyWorkflow:
         materialized, so the key (op name + version + input hashes) is
         always computable, including for chains of cached ops.
         """
+        if getattr(self, "_aborted_reason", None) is not None:
+            # an abort that landed BETWEEN barriers already tore the
+            # runtime down and dropped values: materializations after it
+            # must surface the abort, not a puzzling missing-entry error
+            raise WorkflowAbortedError(
+                f"workflow {self.name} was aborted: {self._aborted_reason}"
+            )
         if not self._call_queue:
             return
         calls = self._call_queue
