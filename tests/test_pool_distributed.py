@@ -190,6 +190,41 @@ def test_worker_death_during_stream_plan(tmp_path):
     assert "PLANDEATH-NOT-DETECTED" not in out0
 
 
+def test_murder_fuzz_pool_survives(tmp_path):
+    """A rank hard-exits mid-campaign; affected seeds fail typed or
+    retry, and every later seed completes on the survivors."""
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    base_env = dict(os.environ)
+    base_env["LZY_AMD_STORAGE"] = str(tmp_path / "storage")
+    base_env["PYTHONPATH"] = str(ROOT) + os.pathsep + base_env.get("PYTHONPATH", "")
+    base_env.update(
+        MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port), WORLD_SIZE="3",
+        FUZZ_MURDER="1:20", FUZZ_ROUNDS="25", FUZZ_BASE_SEED="52000",
+    )
+    procs = []
+    for rank in range(3):
+        env = dict(base_env)
+        env["RANK"] = str(rank)
+        env["LOCAL_RANK"] = str(rank)
+        procs.append(subprocess.Popen(
+            [sys.executable, "tests/pool_script_fuzz.py"],
+            cwd=ROOT, env=env, stdout=subprocess.PIPE, stderr=subprocess.PIPE,
+            text=True,
+        ))
+    out0, err0 = procs[0].communicate(timeout=240)
+    for p in procs[1:]:
+        try:
+            p.wait(timeout=60)
+        except subprocess.TimeoutExpired:
+            p.kill()
+    assert "FUZZ-OK" in out0, out0[-3000:] + err0[-2000:]
+    assert procs[1].returncode == 66  # the injected murder fired
+
+
 def test_live_remote_log_tail(tmp_path):
     """A remote rank's print appears on the driver console BEFORE the op
     finishes (live ReadStdSlots-style streaming, not completion-time)."""
