@@ -203,7 +203,7 @@ def test_murder_fuzz_pool_survives(tmp_path):
     base_env["PYTHONPATH"] = str(ROOT) + os.pathsep + base_env.get("PYTHONPATH", "")
     base_env.update(
         MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port), WORLD_SIZE="3",
-        FUZZ_MURDER="1:20", FUZZ_ROUNDS="25", FUZZ_BASE_SEED="52000",
+        FUZZ_MURDER="1:6", FUZZ_ROUNDS="25", FUZZ_BASE_SEED="52000",
     )
     procs = []
     for rank in range(3):
