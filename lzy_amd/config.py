@@ -47,7 +47,7 @@ class Config:
     channel_chunk_mb: int = 256        # chunk size for large-tensor transfers
     channel_wire_cast: str = ""        # ""|fp16|bf16|fp8e4m3|fp8e5m2: lossy wire dtype
     stream_merge: bool = True          # fold pair_reduce trees into streamed plans
-    stream_chunk_mb: int = 64          # chunk size for streamed tree plans
+    stream_chunk_mb: int = 32          # chunk size for streamed tree plans (1 GiB shard -> 32 chunks; pipeline fill ~2 chunks per extra level)
     # HIP data-plane kernels
     hip_max_blocks: int = 0            # 0 -> kernel default grid cap
     op_streams: int = 4                # HIP streams per device for op overlap
