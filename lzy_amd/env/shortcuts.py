@@ -35,7 +35,12 @@ def env_vars(**variables: str) -> LzyEnvironment:
 def manual_python(python_version: Optional[str] = None,
                   libraries: Optional[dict] = None) -> LzyEnvironment:
     """Reference: pin python version + libs for the remote conda env.
-    Workers share this interpreter; a mismatched pin warns loudly."""
+
+    Single-node semantics: workers share this interpreter, so instead of
+    INSTALLING the request (CondaEnvironment.java:67-134) we VALIDATE
+    it — a library that is not importable here cannot be provisioned and
+    fails fast like an unsatisfiable pool; a version mismatch warns (the
+    env exists but differs from the pin)."""
     if python_version is not None:
         current = f"{sys.version_info.major}.{sys.version_info.minor}"
         if not python_version.startswith(current):
@@ -44,6 +49,25 @@ def manual_python(python_version: Optional[str] = None,
                 f"in-process on python {current} (no env re-provisioning "
                 f"on a single node)"
             )
+    for name, want in (libraries or {}).items():
+        try:
+            from importlib import metadata
+
+            have = metadata.version(name)
+        except Exception:
+            from lzy_amd.exceptions import BadProvisioningError
+
+            raise BadProvisioningError(
+                f"manual_python requires library {name!r}, which is not "
+                "installed in this image — a single-node pool cannot "
+                "install packages (no conda re-provisioning by design)"
+            )
+        if want and str(want) not in (have, ""):
+            if not str(have).startswith(str(want)):
+                warnings.warn(
+                    f"manual_python: {name} pinned to {want} but the "
+                    f"shared interpreter has {have}"
+                )
     return LzyEnvironment()
 
 

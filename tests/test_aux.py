@@ -455,3 +455,23 @@ def test_provisioning_custom_score():
     assert big.gpu_count == 8
     cheap = Provisioning(gpu_count=1).resolve_pool(pools)
     assert cheap.gpu_count == 1
+
+
+def test_manual_python_validates_libraries():
+    """Single-node conda analogue: the library pin is VALIDATED against
+    the shared interpreter (missing -> BadProvisioningError like an
+    unsatisfiable pool; version mismatch -> warning)."""
+    import warnings as _w
+
+    import pytest as _pt
+
+    from lzy_amd.env import shortcuts as sc
+    from lzy_amd.exceptions import BadProvisioningError
+
+    sc.manual_python(libraries={"numpy": ""})  # present: fine
+    with _pt.raises(BadProvisioningError):
+        sc.manual_python(libraries={"surely_not_installed_xyz": "1.0"})
+    with _w.catch_warnings(record=True) as rec:
+        _w.simplefilter("always")
+        sc.manual_python(libraries={"numpy": "0.0"})
+        assert any("pinned to 0.0" in str(x.message) for x in rec)
