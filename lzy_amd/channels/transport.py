@@ -381,11 +381,57 @@ class Transport:
         return ops, (lambda: unpickle_value(buf.numpy().tobytes()))
 
     @staticmethod
+    def _batch(ops: list) -> list:
+        """batch_isend_irecv normalized to ONE work PER OP.
+
+        Critical NCCL/RCCL behavior: with a coalescing backend the call
+        returns a SINGLE work for the whole group (cm.works from
+        _coalescing_manager), not one per op.  Callers here slice works
+        per entry/chunk (settle waits, resume offsets); naive slicing
+        would leave later entries with EMPTY work lists — no stream wait
+        before their buffers are read.  Replicating the group work per
+        op keeps every wait correct (waiting the same work repeatedly is
+        idempotent).  gloo returns per-op works and passes through."""
+        works = dist.batch_isend_irecv(ops)
+        if len(works) == len(ops):
+            return works
+        if len(works) == 1:
+            return [works[0]] * len(ops)
+
+        class _AllOf:  # unexpected shape: conservative wait-them-all
+            def __init__(self, ws):
+                self._ws = ws
+
+            def wait(self, *a, **k):
+                for w in self._ws:
+                    w.wait(*a, **k)
+                return True
+
+            def is_completed(self):
+                return all(w.is_completed() for w in self._ws)
+
+        return [_AllOf(works)] * len(ops)
+
+    @staticmethod
     def issue(ops: list) -> list:
-        """Issue a group of P2POps as one batch; returns the works."""
+        """Issue a group of P2POps; returns one work per op, in order.
+
+        Device and host ops are batched SEPARATELY: on the composite
+        backend (cuda:nccl + cpu:gloo) a mixed batch would coalesce two
+        different backends in one group — splitting is always safe
+        because matching happens per backend, and both endpoints derive
+        the same split from the same item order."""
         if not ops:
             return []
-        return dist.batch_isend_irecv(ops)
+        cuda_ops = [op for op in ops if op.tensor.is_cuda]
+        cpu_ops = [op for op in ops if not op.tensor.is_cuda]
+        if not cuda_ops or not cpu_ops:
+            return Transport._batch(ops)
+        works_by_id = {}
+        for group in (cuda_ops, cpu_ops):
+            for op, w in zip(group, Transport._batch(group)):
+                works_by_id[id(op)] = w
+        return [works_by_id[id(op)] for op in ops]
 
     # -- single-transfer conveniences ----------------------------------------
 

@@ -165,3 +165,52 @@ def test_wirecast_decision_deterministic_and_sound(dtype, numel, wire):
 def test_serializer_bytes_roundtrip(data):
     back = _roundtrip(data)
     assert back == data
+
+
+def test_issue_normalizes_coalesced_works(monkeypatch):
+    """NCCL/RCCL coalescing returns ONE work for a whole batch; issue()
+    must still hand back one (waitable) work per op, or settle slicing
+    would skip stream waits on later entries (silent corruption on the
+    real multi-GPU node)."""
+    import torch.distributed as dist
+
+    from lzy_amd.channels.transport import Transport
+
+    class FakeWork:
+        def __init__(self):
+            self.waits = 0
+
+        def wait(self, *a, **k):
+            self.waits += 1
+            return True
+
+        def is_completed(self):
+            return True
+
+    class FakeOp:
+        class _T:
+            is_cuda = False
+
+        tensor = _T()
+
+    ops = [FakeOp(), FakeOp(), FakeOp()]
+
+    single = FakeWork()
+    monkeypatch.setattr(dist, "batch_isend_irecv", lambda o: [single])
+    works = Transport.issue(list(ops))
+    assert len(works) == 3 and all(w is single for w in works)
+    for w in works:
+        w.wait()
+    assert single.waits == 3  # idempotent repeated waits
+
+    per_op = [FakeWork() for _ in ops]
+    monkeypatch.setattr(dist, "batch_isend_irecv", lambda o: list(per_op))
+    works = Transport.issue(list(ops))
+    assert works == per_op
+
+    two = [FakeWork(), FakeWork()]
+    monkeypatch.setattr(dist, "batch_isend_irecv", lambda o: list(two))
+    works = Transport.issue(list(ops))
+    assert len(works) == 3
+    works[2].wait()
+    assert all(w.waits == 1 for w in two)  # conservative all-of wait
