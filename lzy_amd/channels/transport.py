@@ -312,7 +312,11 @@ class Transport:
                 chunks = [t]
             else:
                 chunks = self._chunks(t.view(-1), per, offset_chunks)
-            ops = [dist.P2POp(dist.isend, c, dst, group=self._pg, tag=tag)
+            # tags are a GLOO matching feature; the NCCL/RCCL path is
+            # matched by issue order and torch documents tags as
+            # unsupported there — pass 0 so no backend can object
+            eff = 0 if t.is_cuda else tag
+            ops = [dist.P2POp(dist.isend, c, dst, group=self._pg, tag=eff)
                    for c in chunks]
             return ops, t
         data = prepickled if prepickled is not None else pickle_value(value)
@@ -358,7 +362,8 @@ class Transport:
                 chunks = [buf]
             else:
                 chunks = self._chunks(buf.view(-1), per, offset_chunks)
-            ops = [dist.P2POp(dist.irecv, c, src, group=self._pg, tag=tag)
+            eff = 0 if buf.is_cuda else tag  # mirror of the sender rule
+            ops = [dist.P2POp(dist.irecv, c, src, group=self._pg, tag=eff)
                    for c in chunks]
 
             def finalize():

@@ -391,6 +391,10 @@ def run_stream_plan(
         nonce = plan.get("nonce", 0)
 
         def edge_tag(topo: int) -> int:
+            # gloo-only matching aid; the RCCL path is order-matched and
+            # torch documents tags as unsupported there
+            if on_device:
+                return 0
             return (nonce + topo) % (1 << 30)
 
         def post_up_to(src: int, upto_flat_idx: int) -> None:
