@@ -46,6 +46,18 @@ def test_pool_two_ranks(tmp_path):
     assert "POOL-SCRIPT-OK" in res.stdout
 
 
+def test_pool_two_ranks_coalesced_works(tmp_path):
+    """Same battery under a simulated NCCL-coalesced batch_isend_irecv
+    (ONE work per group, the shape RCCL returns on the real node):
+    Transport.issue must normalize so settle slicing stays correct."""
+    res = _run_distributed("tests/pool_script_coalesce.py", 2, tmp_path)
+    if res.returncode != 0:
+        print("STDOUT:", res.stdout[-4000:])
+        print("STDERR:", res.stderr[-4000:])
+    assert res.returncode == 0
+    assert "POOL-SCRIPT-OK" in res.stdout
+
+
 def test_pool_single_rank_inprocess(tmp_path, monkeypatch):
     """ws=1 degenerates to driver-only: full pipeline without dist."""
     env = dict(os.environ)
