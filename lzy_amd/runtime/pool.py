@@ -20,7 +20,17 @@ Topology:
     rank issues its p2p ops in a pairwise-consistent order (RCCL matches
     by order, not tags — no deadlocks by construction).
   * gang ops (@op(gpu_count=k)): the same TaskSpec dispatched to k ranks
-    with a shared RCCL subgroup for in-op collectives/DDP.
+    with a shared RCCL subgroup for in-op collectives/DDP; gangs with
+    disjoint rank sets run concurrently.
+  * streamed merge-tree plans (@op(pair_reduce=...)): connected merge
+    components execute as one chunk-pipelined p2p collective on a
+    dedicated process group (channels/treeplan.py).
+  * failure model: mid-flight StopGraph (abort/Ctrl-C cancels queued
+    tasks, bounded drain); worker death re-dispatches inflight tasks to
+    survivors with inputs re-rooted from live owners or the durable
+    tier; transfer-source death fails consumers fast (bounded settle
+    waits, per-transfer gloo tags) and retries through the same
+    machinery.  Dead ranks stay excluded from all later placement.
 """
 from __future__ import annotations
 
