@@ -15,9 +15,11 @@ import torch
 from lzy_amd import ops
 
 VARIANTS_16 = {0: "V16 U4 (cur)", 1: "V16 U8", 2: "V32 U2", 3: "V32 U4",
-               4: "V8 U8", 5: "V16 U2", 6: "V8 U4"}
+               4: "V8 U8", 5: "V16 U2", 6: "V8 U4",
+               7: "V16 U4 NT", 8: "V8 U8 NT", 9: "V16 U2 NT"}
 VARIANTS_32 = {0: "V8 U4 (cur)", 1: "V8 U8", 2: "V16 U2", 3: "V16 U4",
-               4: "V4 U8", 5: "V8 U2", 6: "V4 U4"}
+               4: "V4 U8", 5: "V8 U2", 6: "V4 U4",
+               7: "V8 U4 NT", 8: "V16 U4 NT", 9: "V8 U2 NT"}
 
 
 def bw(t, fn, iters):
