@@ -372,3 +372,23 @@ def test_transport_packs_transposed_view():
     packed = _pack_contiguous(view)
     assert packed.is_contiguous()
     assert torch.equal(packed, view.contiguous())
+
+
+@requires_gpu
+def test_stats_variants_match_reference():
+    """Every compiled (V, U, NT) sweep variant of the stats reduction —
+    including the non-temporal-load variants 7-9 — produces the same
+    sum / sum-of-squares as the plain torch fp64 reference.  Variants
+    stay compiled even though the default shape won the sweep
+    (profiles/stats_sweep_r02_nt.txt); this pins their correctness."""
+    from lzy_amd.ops import stats_variant
+
+    for dtype in (torch.bfloat16, torch.float32):
+        t = torch.randn((1 << 20) + 13, device="cuda", dtype=dtype)
+        ref = t.double()
+        want_s, want_s2 = float(ref.sum()), float((ref * ref).sum())
+        for v in range(10):
+            got = stats_variant(t, v).cpu()
+            tol = 2e-3 if dtype is torch.bfloat16 else 1e-3
+            assert abs(float(got[0]) - want_s) <= tol * max(1.0, abs(want_s)), (dtype, v)
+            assert abs(float(got[1]) - want_s2) <= tol * max(1.0, want_s2), (dtype, v)
