@@ -11,8 +11,21 @@ multi-GPU ops.
 from lzy_amd.core.lzy import Lzy, lzy_auth
 from lzy_amd.core.op import op
 from lzy_amd.core.workflow import LzyWorkflow
+from lzy_amd.env.compat import (
+    AutoPythonEnv,
+    DockerContainer,
+    DockerPullPolicy,
+    ManualPythonEnv,
+    NoContainer,
+)
 from lzy_amd.env.environment import LzyEnvironment
-from lzy_amd.env.provisioning import GpuType, Provisioning
+from lzy_amd.env.provisioning import (
+    Any as AnyProvisioning,
+    GpuType,
+    Provisioning,
+    maximum_score_function,
+    minimum_score_function,
+)
 from lzy_amd.proxy import (
     is_lzy_proxy,
     materialize,
@@ -31,6 +44,14 @@ __all__ = [
     "LzyEnvironment",
     "Provisioning",
     "GpuType",
+    "AnyProvisioning",
+    "DockerContainer",
+    "DockerPullPolicy",
+    "NoContainer",
+    "AutoPythonEnv",
+    "ManualPythonEnv",
+    "maximum_score_function",
+    "minimum_score_function",
     "File",
     "whiteboard_",
     "whiteboard",

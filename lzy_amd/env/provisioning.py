@@ -28,6 +28,44 @@ class GpuType(enum.Enum):
     T4 = "T4"
 
 
+class AnyRequirement:
+    """'Any value is fine' sentinel (reference: provisioning.py:43
+    ``Any = AnyRequirement()``).  Our Provisioning uses None for the
+    same meaning; shortcuts normalize this sentinel to None so
+    reference scripts that pass ``Any`` run unchanged."""
+
+    _instance: Optional["AnyRequirement"] = None
+
+    def __new__(cls) -> "AnyRequirement":
+        if cls._instance is None:
+            cls._instance = super().__new__(cls)
+        return cls._instance
+
+    def __repr__(self) -> str:
+        return "Any"
+
+
+Any = AnyRequirement()
+
+
+def _norm(v):
+    """Map the Any sentinel (and reference NotSpecified-style Nones) to
+    this module's None convention."""
+    return None if isinstance(v, AnyRequirement) else v
+
+
+def maximum_score_function(requested: "Provisioning", pool: "PoolSpec") -> float:
+    """Prefer the LARGEST feasible pool (reference: env/provisioning/
+    score.py:16 — biggest headroom over the request)."""
+    return float(pool.gpu_count * 1024 + pool.cpu_count)
+
+
+def minimum_score_function(requested: "Provisioning", pool: "PoolSpec") -> float:
+    """Prefer the TIGHTEST feasible fit (reference: score.py:28 — least
+    over-allocation; the default ranking in resolve_pool)."""
+    return -float(pool.gpu_count * 1024 + pool.cpu_count)
+
+
 @dataclass(frozen=True)
 class PoolSpec:
     """One placement class offered by the node."""

@@ -85,3 +85,73 @@ def docker_image(image: str, pull_policy: str = "IF_NOT_EXISTS") -> LzyEnvironme
         f"(no container runtime in the data plane)"
     )
     return LzyEnvironment()
+
+
+# ---------------------------------------------------------------------------
+# Reference-named shortcut spellings (pylzy/lzy/env/shortcuts.py:28-103).
+# The reference vends appliers that mutate a decorated subject; here every
+# shortcut returns an LzyEnvironment for `@op(env=...)` / `.with_env(...)`
+# (see docs/migrating.md) — same names, same keyword signatures.
+# ---------------------------------------------------------------------------
+
+
+def docker_container(*, registry: str, image: str,
+                     pull_policy=None, username: Optional[str] = None,
+                     password: Optional[str] = None) -> LzyEnvironment:
+    from lzy_amd.env.compat import DockerContainer, DockerPullPolicy
+
+    DockerContainer(
+        registry=registry, image=image,
+        pull_policy=pull_policy or DockerPullPolicy.IF_NOT_EXISTS,
+        username=username, password=password,
+    ).validate()
+    return LzyEnvironment()
+
+
+def no_container() -> LzyEnvironment:
+    """In-process execution IS the no-container mode."""
+    return LzyEnvironment()
+
+
+def manual_python_env(*, python_version: str,
+                      local_module_paths: Optional[Sequence[str]] = None,
+                      pypi_packages: Optional[dict] = None,
+                      pypi_index_url: Optional[str] = None) -> LzyEnvironment:
+    from lzy_amd.env.compat import ManualPythonEnv
+
+    ManualPythonEnv(
+        python_version=python_version,
+        local_module_paths=tuple(local_module_paths or ()),
+        pypi_packages=dict(pypi_packages or {}),
+        pypi_index_url=pypi_index_url,
+    ).validate()
+    return LzyEnvironment()
+
+
+def auto_python_env(*, pypi_index_url: Optional[str] = None,
+                    additional_pypi_packages: Optional[dict] = None) -> LzyEnvironment:
+    from lzy_amd.env.compat import AutoPythonEnv
+
+    AutoPythonEnv(
+        pypi_index_url=pypi_index_url,
+        additional_pypi_packages=additional_pypi_packages,
+    ).validate()
+    return LzyEnvironment()
+
+
+def provisioning(*, cpu_type=None, cpu_count=None, gpu_type=None,
+                 gpu_count=None, ram_size_gb=None) -> LzyEnvironment:
+    """Reference keyword spelling (`Any` sentinel accepted for every
+    field; cpu_type has no meaning on a homogeneous node)."""
+    from lzy_amd.env.provisioning import Provisioning, _norm
+
+    if isinstance(gpu_type, object) and gpu_type is not None:
+        gpu_type = _norm(gpu_type)
+        if hasattr(gpu_type, "value"):  # GpuType enum member
+            gpu_type = gpu_type.value
+    return LzyEnvironment(provisioning=Provisioning(
+        cpu_count=_norm(cpu_count),
+        ram_size_gb=_norm(ram_size_gb),
+        gpu_count=_norm(gpu_count),
+        gpu_type=gpu_type,
+    ))

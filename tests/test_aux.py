@@ -475,3 +475,56 @@ def test_manual_python_validates_libraries():
         _w.simplefilter("always")
         sc.manual_python(libraries={"numpy": "0.0"})
         assert any("pinned to 0.0" in str(x.message) for x in rec)
+
+
+def test_reference_named_env_compat():
+    """The reference's public env names (pylzy/lzy/api/v1/__init__.py:5-27)
+    import and behave: containers validate-and-warn, python envs validate
+    the shared interpreter, Any normalizes to 'unconstrained'."""
+    import warnings
+
+    import pytest
+
+    from lzy_amd import (
+        AnyProvisioning, AutoPythonEnv, DockerContainer, DockerPullPolicy,
+        ManualPythonEnv, NoContainer, maximum_score_function,
+        minimum_score_function,
+    )
+    from lzy_amd.env.provisioning import PoolSpec, Provisioning
+    from lzy_amd.env.shortcuts import (
+        auto_python_env, docker_container, manual_python_env, no_container,
+        provisioning,
+    )
+    from lzy_amd.exceptions import BadProvisioningError
+
+    assert NoContainer().deconstruct() == {}
+    with warnings.catch_warnings(record=True) as w:
+        warnings.simplefilter("always")
+        docker_container(registry="r.io", image="img:1",
+                         pull_policy=DockerPullPolicy.ALWAYS)
+        assert any("ignored" in str(x.message) for x in w)
+    assert DockerContainer(registry="r", image="i").get_image() == "i"
+
+    assert no_container().provisioning == Provisioning()
+    assert auto_python_env().env_variables == {}
+    manual_python_env(python_version="3.10", pypi_packages={"numpy": ""})
+    with pytest.raises(BadProvisioningError):
+        manual_python_env(python_version="3.10",
+                          pypi_packages={"surely_not_installed_xyz": "1"})
+    with pytest.raises(BadProvisioningError):
+        AutoPythonEnv(
+            additional_pypi_packages={"surely_not_installed_xyz": "1"}
+        ).validate()
+
+    # Any sentinel is a singleton and normalizes to unconstrained
+    assert AnyProvisioning is type(AnyProvisioning)()
+    env = provisioning(gpu_count=2, cpu_count=AnyProvisioning,
+                       ram_size_gb=AnyProvisioning)
+    assert env.provisioning == Provisioning(gpu_count=2)
+
+    # score functions rank pools opposite ways
+    small = PoolSpec("s", 8, 64, 1, "MI355X")
+    big = PoolSpec("b", 64, 512, 8, "MI355X")
+    p = Provisioning(gpu_count=1)
+    assert p.resolve_pool([small, big], score=maximum_score_function) is big
+    assert p.resolve_pool([small, big], score=minimum_score_function) is small
