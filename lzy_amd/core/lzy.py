@@ -134,6 +134,12 @@ class Lzy(WithEnvironmentMixin):
         root.mkdir(parents=True, exist_ok=True)
         return WhiteboardIndexClient(str(root / "whiteboards.db"))
 
+    @property
+    def whiteboard_manager(self) -> WhiteboardIndexClient:
+        """Reference-named alias (pylzy/lzy/core/lzy.py:85
+        ``whiteboard_manager``) for the whiteboard index."""
+        return self.whiteboard_index
+
     # -- workflows ----------------------------------------------------------
 
     def workflow(
