@@ -43,6 +43,19 @@ class LzyEnvironment:
             namespace={**self.namespace, **other.namespace},
         )
 
+    def __call__(self, subject):
+        """Reference applier spelling (pylzy/lzy/env/shortcuts.py:28-103
+        EnvironmentApplierType): shortcuts stack as decorators ABOVE
+        @op — ``@provisioning(cpu_count=8)`` / ``@gpu(2)`` over an op
+        wrapper merges this env into it (this env's fields win)."""
+        if hasattr(subject, "with_env") and hasattr(subject, "env"):
+            return subject.with_env(subject.env.combine(self))
+        raise TypeError(
+            "an LzyEnvironment shortcut used as a decorator must wrap an "
+            "@op wrapper (or another env-bearing object) — put it ABOVE "
+            "@op, e.g.\n  @provisioning(cpu_count=8)\n  @op\n  def f(): ..."
+        )
+
 
 class WithEnvironmentMixin:
     """Fluent env shortcuts shared by Lzy / workflow / op wrappers

@@ -528,3 +528,48 @@ def test_reference_named_env_compat():
     p = Provisioning(gpu_count=1)
     assert p.resolve_pool([small, big], score=maximum_score_function) is big
     assert p.resolve_pool([small, big], score=minimum_score_function) is small
+
+
+def test_env_shortcut_applier_decorators():
+    """The reference stacks env shortcuts as decorators ABOVE @op
+    (pylzy/examples/provisioning.py) — the returned LzyEnvironment is a
+    callable applier that merges into the op wrapper."""
+    import pytest
+
+    from lzy_amd import AnyProvisioning, Lzy, op
+    from lzy_amd.env.shortcuts import env_vars, gpu, provisioning
+
+    @provisioning(cpu_count=8)
+    @op
+    def two() -> int:
+        return 2
+
+    @provisioning(gpu_count=AnyProvisioning)
+    @op
+    def three() -> int:
+        return 3
+
+    @gpu(2)
+    @op
+    def four() -> int:
+        return 4
+
+    assert two.env.provisioning.cpu_count == 8
+    assert three.env.provisioning.gpu_count is None
+    assert four.env.provisioning.gpu_count == 2
+    assert two() == 2  # outside a workflow: plain function
+
+    @env_vars(LZY_TEST_APPLIER="on")
+    @op
+    def probe() -> str:
+        import os
+        return os.environ.get("LZY_TEST_APPLIER", "")
+
+    with Lzy().workflow("applier_env") as wf:
+        got = str(probe())
+    assert got == "on"
+
+    with pytest.raises(TypeError):
+        @provisioning(cpu_count=1)
+        def raw() -> None:
+            ...
