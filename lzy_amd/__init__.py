@@ -26,6 +26,7 @@ from lzy_amd.env.provisioning import (
     maximum_score_function,
     minimum_score_function,
 )
+from lzy_amd.env import score
 from lzy_amd.proxy import (
     is_lzy_proxy,
     materialize,
@@ -52,6 +53,7 @@ __all__ = [
     "ManualPythonEnv",
     "maximum_score_function",
     "minimum_score_function",
+    "score",
     "File",
     "whiteboard_",
     "whiteboard",
