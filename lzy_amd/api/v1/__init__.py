@@ -37,12 +37,16 @@ from lzy_amd.env.shortcuts import (
 from lzy_amd.runtime.base import Runtime
 from lzy_amd.runtime.local import LocalRuntime
 from lzy_amd.runtime.pool import GpuPoolRuntime
+
+# Reference-named alias: the "remote" runtime of this node IS the GPU
+# pool (reference: lzy/api/v1/remote/runtime.py RemoteRuntime).
+RemoteRuntime = GpuPoolRuntime
 from lzy_amd.types import File
 from lzy_amd.whiteboards.wb import whiteboard, whiteboard_
 
 __all__ = [
     "Lzy", "lzy_auth", "op", "LzyWorkflow", "LzyEnvironment",
-    "Provisioning", "GpuType", "Runtime", "LocalRuntime", "GpuPoolRuntime", "File",
+    "Provisioning", "GpuType", "Runtime", "LocalRuntime", "GpuPoolRuntime", "RemoteRuntime", "File",
     "whiteboard", "whiteboard_", "gpu", "cpu", "ram", "env_vars",
     "auto_python", "manual_python", "docker_image",
     "AnyProvisioning", "DockerContainer", "DockerPullPolicy",
