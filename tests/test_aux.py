@@ -573,3 +573,22 @@ def test_env_shortcut_applier_decorators():
         @provisioning(cpu_count=1)
         def raw() -> None:
             ...
+
+
+def test_reference_api_v1_export_list_resolves():
+    """Every name the reference exports from lzy.api.v1
+    (pylzy/lzy/api/v1/__init__.py:5-40) resolves from lzy_amd.api.v1 —
+    a reference script migrates with only the package rename."""
+    import lzy_amd.api.v1 as v1
+
+    ref_exports = [
+        "Runtime", "LocalRuntime", "RemoteRuntime", "op", "Lzy",
+        "lzy_auth", "LzyEnvironment", "DockerContainer",
+        "DockerPullPolicy", "NoContainer", "score", "Provisioning",
+        "AnyProvisioning", "AutoPythonEnv", "ManualPythonEnv",
+        "docker_container", "no_container", "provisioning",
+        "auto_python_env", "manual_python_env", "env_vars", "whiteboard",
+    ]
+    missing = [n for n in ref_exports if not hasattr(v1, n)]
+    assert not missing, missing
+    assert v1.RemoteRuntime is v1.GpuPoolRuntime
