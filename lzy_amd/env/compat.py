@@ -94,6 +94,11 @@ class ManualPythonEnv:
                     f"ManualPythonEnv: local module path {path!r} does not "
                     "exist on this node"
                 )
+        # reference flag (pylzy/lzy/config.py:34 LZY_SKIP_PYPI_VALIDATION):
+        # opt out of package-pin validation entirely
+        if os.environ.get("LZY_SKIP_PYPI_VALIDATION", "").lower() in (
+                "1", "yes", "true", "on"):
+            return
         from importlib import metadata
 
         for name, want in self.pypi_packages.items():

@@ -592,3 +592,13 @@ def test_reference_api_v1_export_list_resolves():
     missing = [n for n in ref_exports if not hasattr(v1, n)]
     assert not missing, missing
     assert v1.RemoteRuntime is v1.GpuPoolRuntime
+
+
+def test_skip_pypi_validation_flag(monkeypatch):
+    """LZY_SKIP_PYPI_VALIDATION (reference: pylzy/lzy/config.py:34)
+    bypasses package-pin validation."""
+    from lzy_amd.env.compat import ManualPythonEnv
+
+    monkeypatch.setenv("LZY_SKIP_PYPI_VALIDATION", "true")
+    ManualPythonEnv(python_version="3.10",
+                    pypi_packages={"surely_not_installed_xyz": "1"}).validate()
