@@ -602,3 +602,40 @@ def test_skip_pypi_validation_flag(monkeypatch):
     monkeypatch.setenv("LZY_SKIP_PYPI_VALIDATION", "true")
     ManualPythonEnv(python_version="3.10",
                     pypi_packages={"surely_not_installed_xyz": "1"}).validate()
+
+
+def test_logging_config_surface(monkeypatch, capsys, tmp_path):
+    """configure_logging / get_logger / get_remote_logger honor
+    LZY_LOG_LEVEL and LZY_LOG_CONFIG_PATH (reference:
+    pylzy/lzy/logs/config.py:32-66)."""
+    import json
+    import logging
+
+    from lzy_amd.utils.logconfig import (
+        configure_logging, get_logger, get_remote_logger,
+        get_logging_config,
+    )
+
+    monkeypatch.setenv("LZY_LOG_LEVEL", "debug")
+    cfg = get_logging_config()
+    assert cfg["loggers"]["lzy"]["level"] == "DEBUG"
+    configure_logging(cfg)
+    lg = get_logger("scheduler")
+    assert lg.name == "lzy.scheduler"
+    assert get_logger("lzy.scheduler") is logging.getLogger("lzy.scheduler")
+    assert get_remote_logger("rank0").name == "remote.rank0"
+    assert lg.isEnabledFor(logging.DEBUG)
+
+    p = tmp_path / "log.json"
+    p.write_text(json.dumps({
+        "version": 1, "disable_existing_loggers": False,
+        "loggers": {"lzy": {"level": "ERROR"}},
+    }))
+    monkeypatch.setenv("LZY_LOG_CONFIG_PATH", str(p))
+    assert get_logging_config()["loggers"]["lzy"]["level"] == "ERROR"
+    configure_logging()
+    assert not get_logger("x").isEnabledFor(logging.INFO)
+    # restore a sane config for later tests
+    monkeypatch.delenv("LZY_LOG_CONFIG_PATH")
+    monkeypatch.setenv("LZY_LOG_LEVEL", "INFO")
+    configure_logging()
