@@ -186,3 +186,100 @@ def test_build_plan_chain_shape():
     srcs = [s["remote_src"] for s in plan["steps_by_rank"][0]
             if s["op"] == "node"]
     assert srcs == [1, 2, 3]
+
+
+# ---------------------------------------------------------------------------
+# property-based: random DAGs / random trees (hypothesis)
+# ---------------------------------------------------------------------------
+from hypothesis import given, settings, strategies as st
+
+
+@settings(max_examples=80, deadline=None)
+@given(st.data())
+def test_find_components_invariants_random_dags(data):
+    """For ARBITRARY mixes of pair-reduce and disqualified ops:
+    components are disjoint, every member qualifies (pair_reduce,
+    2 inputs, 1 output, no kwargs, not cached), each component has >=2
+    members, and member order is topological."""
+    FakeCall._n = 0
+    n_leaves = data.draw(st.integers(2, 10), label="leaves")
+    avail = [f"L{i}" for i in range(n_leaves)]
+    calls = {}
+    for k in range(data.draw(st.integers(0, 14), label="ops")):
+        ins = data.draw(
+            st.lists(st.sampled_from(avail), min_size=2, max_size=3),
+            label=f"in{k}")
+        kind = data.draw(st.sampled_from(
+            ["pair", "pair", "pair", "plain", "cached", "kw", "multi"]),
+            label=f"kind{k}")
+        c = FakeCall(
+            ins[:2] if kind != "pair" or len(ins) < 3 else ins[:2],
+            outputs=2 if kind == "multi" else 1,
+            pair_reduce=(1.0, 1.0) if kind != "plain" else None,
+            kwargs={"k": ins[-1]} if kind == "kw" else None,
+        )
+        if kind == "cached":
+            c.cache = True
+        calls[f"t{k}"] = c
+        avail.extend(c.entry_ids)
+
+    comps = find_components(calls)
+    seen = set()
+    for comp in comps:
+        assert len(comp) >= 2
+        assert not (set(comp) & seen)  # disjoint
+        seen |= set(comp)
+        pos = {t: i for i, t in enumerate(comp)}
+        out_of = {}
+        for t in comp:
+            c = calls[t]
+            assert c.pair_reduce is not None
+            assert len(c.entry_ids) == 1
+            assert not c.kwarg_entry_ids
+            assert not getattr(c, "cache", False)
+            assert len(c.input_entry_ids()) == 2
+            out_of[c.entry_ids[0]] = t
+        for t in comp:  # topo order within the component
+            for e in calls[t].input_entry_ids():
+                if e in out_of:
+                    assert pos[out_of[e]] < pos[t]
+
+
+@settings(max_examples=60, deadline=None)
+@given(
+    width=st.integers(3, 16),
+    world=st.integers(1, 8),
+    chunk=st.sampled_from([256, 1024, 4096, 1 << 20]),
+    data=st.data(),
+)
+def test_build_plan_invariants_random_trees(width, world, chunk, data):
+    """Plans built from random widths/placements/chunk sizes are
+    structurally sound: every member gets a rank inside the
+    participant set, every send targets a participant (never itself),
+    and chunking covers the tensor in >=1 pieces (>=4 when it fits)."""
+    owners = [data.draw(st.integers(0, world - 1), label=f"o{i}")
+              for i in range(width)]
+    calls, metas = make_tree(width, owners=owners)
+    comps = find_components(calls)
+    if not comps:  # width<=2 never reaches here (width>=3)
+        return
+    [order] = comps
+    plan = build_plan("pp", order, calls, metas.get,
+                      chunk_bytes=chunk, cuda_p2p=False)
+    if plan is None:  # builder may refuse (e.g. degenerate placement)
+        return
+    parts = set(plan["participants"])
+    assert set(owners) <= parts
+    nr = plan["node_rank"]
+    assert set(nr) == set(order)
+    assert all(r in parts for r in nr.values())
+    n_chunks = -(-plan["numel"] // plan["chunk_elems"])
+    assert n_chunks >= 1
+    if plan["numel"] >= 4:
+        assert n_chunks >= 4 or plan["chunk_elems"] * 4 > plan["numel"]
+    for rank, steps in plan["steps_by_rank"].items():
+        assert rank in parts
+        for s in steps:
+            for d, _topo in s.get("send_to", []):
+                assert d in parts
+                assert d != rank  # no self-sends in the schedule
