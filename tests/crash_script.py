@@ -43,7 +43,13 @@ def stage3(x: int) -> int:
 
 
 def main() -> None:
-    if os.environ.get("CRASH") == "1":
+    at = os.environ.get("CRASH_AT")
+    if at is not None:
+        # die after the (at+1)-th EXECUTED op — cached ops do not count,
+        # so repeated runs with CRASH_AT=0 walk the crash point forward
+        # one stage per restart (reference stress-runner restart chain)
+        FAULTS.arm("executor.after_run", countdown=int(at), kind="exit")
+    elif os.environ.get("CRASH") == "1":
         # kill the process right after the second op persists its result
         FAULTS.arm("executor.after_run", countdown=1, kind="exit")
 

@@ -115,3 +115,38 @@ def test_pool_crash_then_resume(tmp_path):
     assert "RESULT=15" in r2.stdout and "POOL-RESUME-OK" in r2.stdout
     # stage1/2 served from cache: no second execution
     assert count("stage1") == 1 and count("stage2") == 1 and count("stage3") == 1
+
+
+def test_repeated_crash_resume_chain(tmp_path):
+    """Reference stress-runner semantics: kill, restart, kill again,
+    restart, complete — each restart resumes from the persisted results
+    and the crash point walks forward one stage per run."""
+    env = dict(os.environ)
+    env["LZY_AMD_STORAGE"] = str(tmp_path / "storage")
+    env["MARK_DIR"] = str(tmp_path / "marks")
+    env["PYTHONPATH"] = str(ROOT) + os.pathsep + env.get("PYTHONPATH", "")
+    os.makedirs(env["MARK_DIR"], exist_ok=True)
+
+    def run(crash_at):
+        e = dict(env)
+        if crash_at is not None:
+            e["CRASH_AT"] = str(crash_at)
+        return subprocess.run(
+            [sys.executable, "tests/crash_script.py"],
+            cwd=ROOT, env=e, capture_output=True, text=True, timeout=120,
+        )
+
+    # run 1: dies right after the first executed op (stage1)
+    assert run(0).returncode == 42
+    assert _count(tmp_path, "stage1") == 1
+    assert _count(tmp_path, "stage2") == 0
+    # run 2: stage1 cached; dies after the next executed op (stage2)
+    assert run(0).returncode == 42
+    assert _count(tmp_path, "stage1") == 1  # not re-executed
+    assert _count(tmp_path, "stage2") == 1
+    assert _count(tmp_path, "stage3") == 0
+    # run 3: completes; only stage3 executes
+    r = run(None)
+    assert r.returncode == 0 and "CRASH-SCRIPT-DONE" in r.stdout
+    assert (_count(tmp_path, "stage1"), _count(tmp_path, "stage2"),
+            _count(tmp_path, "stage3")) == (1, 1, 1)
